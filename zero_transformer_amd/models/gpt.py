@@ -1,0 +1,172 @@
+"""GPT-2-style decoder-only transformer with ALiBi, built MI355X-first.
+
+Same capabilities as the reference Flax model (src/models/GPT.py:53-113,
+src/models/layers.py:47-191): bias-free q/k/v/out projections, 4x GELU MLP,
+bias-free LayerNorm, weight-tied token embedding / LM head, ALiBi attention
+bias, shifted fp32 cross-entropy loss, 0.02-normal init with 1/sqrt(2N)
+scaling on residual-out projections.
+
+Parameter names follow the torch_compatability .pth contract
+(reference flax_to_pytorch.py:10-35,96-114):
+    wte.weight, norm.weight, lm_head.weight,
+    blocks.{i}.ln1.weight, blocks.{i}.ln2.weight,
+    blocks.{i}.attn.{query,key,value,fc_resid}.weight,
+    blocks.{i}.mlp.{fc1,fc_resid}.weight
+
+The hot ops route through zero_transformer_amd.ops: fused HIP/CDNA4 kernels
+on GPU, fp32 torch reference on CPU.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple, Union
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import ops
+from ..utils.config import load_config
+
+
+class LayerNorm(nn.Module):
+    """Bias-free LayerNorm (flax nn.LayerNorm(use_bias=False), eps 1e-6)."""
+
+    def __init__(self, dim: int, eps: float = 1e-6):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.layer_norm(x, self.weight, self.eps)
+
+
+class CausalSelfAttention(nn.Module):
+    """Causal multi-head attention with optional ALiBi position biasing."""
+
+    def __init__(self, cfg):
+        super().__init__()
+        dim, heads = cfg.embedding_dim, cfg.num_head
+        assert dim % heads == 0
+        self.num_head = heads
+        self.head_dim = dim // heads
+        self.dropout_p = cfg.dropout
+        self.query = nn.Linear(dim, dim, bias=False)
+        self.key = nn.Linear(dim, dim, bias=False)
+        self.value = nn.Linear(dim, dim, bias=False)
+        self.fc_resid = nn.Linear(dim, dim, bias=False)
+        if cfg.alibi_attn:
+            self.register_buffer("slopes", ops.alibi_slopes(heads), persistent=False)
+        else:
+            self.slopes = None
+        self.resid_dropout = nn.Dropout(cfg.dropout)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        B, T, C = x.shape
+        H, D = self.num_head, self.head_dim
+        q = self.query(x).view(B, T, H, D).transpose(1, 2)
+        k = self.key(x).view(B, T, H, D).transpose(1, 2)
+        v = self.value(x).view(B, T, H, D).transpose(1, 2)
+        o = ops.attention(
+            q, k, v, self.slopes, dropout_p=self.dropout_p, training=self.training
+        )
+        o = o.transpose(1, 2).reshape(B, T, C)
+        return self.resid_dropout(self.fc_resid(o))
+
+
+class MLP(nn.Module):
+    """4x expansion GELU MLP (reference layers.py:47-77)."""
+
+    def __init__(self, cfg):
+        super().__init__()
+        dim = cfg.embedding_dim
+        self.fc1 = nn.Linear(dim, 4 * dim, bias=False)
+        self.fc_resid = nn.Linear(4 * dim, dim, bias=False)
+        self.dropout = nn.Dropout(cfg.dropout)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.dropout(self.fc_resid(ops.gelu(self.fc1(x))))
+
+
+class Block(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.ln1 = LayerNorm(cfg.embedding_dim)
+        self.attn = CausalSelfAttention(cfg)
+        self.ln2 = LayerNorm(cfg.embedding_dim)
+        self.mlp = MLP(cfg)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = x + self.attn(self.ln1(x))
+        x = x + self.mlp(self.ln2(x))
+        return x
+
+
+class GPT(nn.Module):
+    """Decoder-only transformer with tied wte/lm_head."""
+
+    def __init__(self, cfg):
+        super().__init__()
+        self.cfg = cfg
+        self.N = cfg.N
+        self.vocab_size = cfg.vocab_size
+        self.block_size = cfg.block_size
+        self.wte = nn.Embedding(cfg.vocab_size, cfg.embedding_dim)
+        self.blocks = nn.ModuleList([Block(cfg) for _ in range(cfg.N)])
+        self.norm = LayerNorm(cfg.embedding_dim)
+        self.lm_head = nn.Linear(cfg.embedding_dim, cfg.vocab_size, bias=False)
+        self.lm_head.weight = self.wte.weight  # weight tying (GPT.py:100)
+        self._init_weights()
+
+    def _init_weights(self):
+        scaled = 0.02 / math.sqrt(2 * self.N)
+        for name, p in self.named_parameters():
+            if p.dim() == 1:  # LayerNorm weights
+                nn.init.ones_(p)
+            elif "fc_resid" in name:  # residual-out projections (layers.py:72,184)
+                nn.init.normal_(p, std=scaled)
+            else:
+                nn.init.normal_(p, std=0.02)
+
+    def forward(
+        self,
+        idx: torch.Tensor,
+        labels: Optional[torch.Tensor] = None,
+    ) -> Union[torch.Tensor, Tuple[torch.Tensor, torch.Tensor]]:
+        x = self.wte(idx)
+        for block in self.blocks:
+            x = block(x)
+        x = self.norm(x)
+        logits = F.linear(x, self.lm_head.weight)
+        if labels is None:
+            return logits
+        # Shifted CE (GPT.py:105-111): predict token t+1 from position t.
+        tgt = labels[..., 1:].reshape(-1)
+        lg = logits[..., :-1, :].reshape(-1, logits.shape[-1])
+        loss = ops.cross_entropy(lg, tgt)
+        return logits, loss
+
+    def num_params(self, non_embedding: bool = False) -> int:
+        n = sum(p.numel() for p in self.parameters())
+        if non_embedding:
+            n -= self.wte.weight.numel()
+        return n
+
+
+def model_getter(
+    model_size: str,
+    config_path: str = "conf/model_config.yaml",
+    return_cfg: bool = False,
+    dtype: torch.dtype = torch.float32,
+):
+    """Build a GPT from a named YAML section (reference GPT.py:116-137)."""
+    configs = load_config(config_path)
+    assert model_size in configs, "Invalid model name provided"
+    assert dtype in (torch.float16, torch.bfloat16, torch.float32), "Invalid dtype provided"
+    model = GPT(configs[model_size])
+    if dtype != torch.float32:
+        model = model.to(dtype)
+    if return_cfg:
+        return model, configs[model_size]
+    return model

@@ -1,0 +1,234 @@
+"""Op dispatch: hand-written CDNA4 HIP kernels on GPU, torch reference on CPU.
+
+Policy: for CUDA (= ROCm/HIP) tensors the in-tree extension `_zta_hip` MUST be
+present — a missing extension raises instead of silently falling back to
+eager PyTorch, so GPU runs always exercise the native kernels. CPU tensors
+use the fp32 reference implementations in ops/reference.py.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from . import reference
+from .reference import alibi_slopes  # re-export
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _try_load_extension():
+    global _EXT, _EXT_ERR
+    if _EXT is not None:
+        return _EXT
+    try:
+        from . import _zta_hip  # in-tree built .so (ops/setup.py build_ext --inplace)
+
+        _EXT = _zta_hip
+    except ImportError as e:  # pragma: no cover - exercised only on GPU boxes
+        _EXT_ERR = str(e)
+        _EXT = None
+    return _EXT
+
+
+def hip_ops():
+    """Return the HIP extension module, raising loudly if unavailable."""
+    ext = _try_load_extension()
+    if ext is None:
+        raise RuntimeError(
+            "zero_transformer_amd HIP extension (_zta_hip) is not built but a "
+            "CUDA tensor reached the ops layer. Build it in-tree with "
+            "`python -m zero_transformer_amd.ops.build` (hipcc, gfx950). "
+            f"Import error: {_EXT_ERR}"
+        )
+    return ext
+
+
+def hip_available() -> bool:
+    return _try_load_extension() is not None
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm (bias-free)
+# ---------------------------------------------------------------------------
+
+
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        ext = hip_ops()
+        x = x.contiguous()
+        y, rstd, mean = ext.layernorm_fwd(x, weight, eps)
+        ctx.save_for_backward(x, weight, rstd, mean)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, rstd, mean = ctx.saved_tensors
+        dx, dw = hip_ops().layernorm_bwd(dy.contiguous(), x, weight, rstd, mean)
+        return dx, dw, None
+
+
+def layer_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    if x.is_cuda:
+        return _LayerNormFn.apply(x, weight, eps)
+    return reference.layer_norm(x, weight, eps)
+
+
+# ---------------------------------------------------------------------------
+# GELU (tanh approximation)
+# ---------------------------------------------------------------------------
+
+
+class _GeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        ctx.save_for_backward(x)
+        return hip_ops().gelu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        return hip_ops().gelu_bwd(dy.contiguous(), x)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        return _GeluFn.apply(x)
+    return reference.gelu(x)
+
+
+# ---------------------------------------------------------------------------
+# Fused causal ALiBi flash attention
+# ---------------------------------------------------------------------------
+
+
+class _FlashAttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, slopes, dropout_p, training):
+        ext = hip_ops()
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        if slopes is None:
+            slopes = torch.zeros(q.shape[1], dtype=torch.float32, device=q.device)
+        if dropout_p > 0.0 and training:
+            seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+        else:
+            seed, dropout_p = 0, 0.0
+        o, lse = ext.attn_fwd(q, k, v, slopes, float(dropout_p), seed)
+        ctx.save_for_backward(q, k, v, slopes, o, lse)
+        ctx.dropout_p = dropout_p
+        ctx.seed = seed
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, slopes, o, lse = ctx.saved_tensors
+        dq, dk, dv = hip_ops().attn_bwd(
+            do.contiguous(), q, k, v, slopes, o, lse, ctx.dropout_p, ctx.seed
+        )
+        return dq, dk, dv, None, None, None
+
+
+def attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    slopes: Optional[torch.Tensor] = None,
+    dropout_p: float = 0.0,
+    training: bool = False,
+    impl: str = "auto",
+) -> torch.Tensor:
+    """Causal (ALiBi-biased) attention. q,k,v: (B, H, T, D) -> (B, H, T, D).
+
+    impl: "auto" (fused on GPU, eager on CPU), "fused", or "eager".
+    """
+    if impl == "auto":
+        impl = "fused" if q.is_cuda else "eager"
+    if impl == "fused":
+        return _FlashAttentionFn.apply(q, k, v, slopes, dropout_p, training)
+    return reference.attention(q, k, v, slopes, dropout_p, training)
+
+
+# ---------------------------------------------------------------------------
+# Fused cross entropy (gather-based, never one-hot)
+# ---------------------------------------------------------------------------
+
+
+class _CrossEntropyFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets):
+        logits = logits.contiguous()
+        loss, lse = hip_ops().cross_entropy_fwd(logits, targets)
+        ctx.save_for_backward(logits, targets, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, targets, lse = ctx.saved_tensors
+        dlogits = hip_ops().cross_entropy_bwd(logits, targets, lse, dloss.contiguous())
+        return dlogits, None
+
+
+def cross_entropy(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+    """Mean CE over rows; logits (N, V) any float dtype, targets (N,) int64."""
+    if logits.is_cuda:
+        return _CrossEntropyFn.apply(logits, targets)
+    return reference.cross_entropy(logits, targets)
+
+
+# ---------------------------------------------------------------------------
+# Fused AdamW on a flat fp32 master shard (ZeRO-1 update)
+# ---------------------------------------------------------------------------
+
+
+def adamw_step(
+    param_f32: torch.Tensor,
+    param_bf16_out: Optional[torch.Tensor],
+    grad: torch.Tensor,
+    exp_avg: torch.Tensor,
+    exp_avg_sq: torch.Tensor,
+    step: int,
+    lr: float,
+    beta1: float = 0.9,
+    beta2: float = 0.95,
+    eps: float = 1e-8,
+    weight_decay: float = 0.0,
+    clip_value: float = 1.0,
+    grad_scale: float = 1.0,
+) -> None:
+    """One fused AdamW step on a flat shard; optionally emits bf16 params.
+
+    Reference semantics (optax chain, main_zero.py:160-168): grad * grad_scale
+    (the 1/accum of xmap_train_functions.py:81), element-wise clip to
+    +-clip_value, Adam moments with bias correction, decoupled weight decay
+    (0 for no-decay buckets), -lr scale. When `param_bf16_out` is given the
+    updated fp32 params are also written as bf16 (the working copy that gets
+    all-gathered).
+    """
+    if param_f32.is_cuda:
+        hip_ops().adamw_step(
+            param_f32,
+            param_bf16_out if param_bf16_out is not None else param_f32.new_empty(0).to(torch.bfloat16),
+            grad,
+            exp_avg,
+            exp_avg_sq,
+            int(step),
+            float(lr),
+            float(beta1),
+            float(beta2),
+            float(eps),
+            float(weight_decay),
+            float(clip_value),
+            float(grad_scale),
+        )
+    else:
+        reference.adamw_update(
+            param_f32, grad, exp_avg, exp_avg_sq, step, lr, beta1, beta2, eps,
+            weight_decay, clip_value, grad_scale,
+        )
+        if param_bf16_out is not None:
+            param_bf16_out.copy_(param_f32.to(torch.bfloat16))

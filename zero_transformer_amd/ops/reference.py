@@ -1,0 +1,135 @@
+"""Pure-PyTorch fp32-reference implementations of every fused HIP op.
+
+These serve two purposes:
+  1. CPU execution path (tests, gloo-based distributed plumbing tests).
+  2. Numerics references for HIP-kernel parity tests (tests/test_ops_gpu.py):
+     each HIP kernel is compared against the plain fp32 PyTorch op here.
+
+Semantics mirror the JAX reference:
+  - fp32 softmax in attention (reference src/models/layers.py:167-173 — bf16
+    softmax caused a documented model failure, logs/580.md:94-98).
+  - tanh-approximate GELU (flax nn.gelu default, reference layers.py:68).
+  - one-hot-free shifted cross entropy in fp32 (reference src/utils/losses.py
+    computes one-hot x log_softmax; we fuse the gather instead).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+
+def alibi_slopes(num_heads: int) -> torch.Tensor:
+    """Standard ALiBi per-head slopes (Press et al., Train Short Test Long).
+
+    Matches reference src/models/layers.py:17-30 (get_slopes).
+    """
+
+    def pow2_slopes(n: int):
+        start = 2.0 ** (-(2.0 ** -(math.log2(n) - 3)))
+        return [start * (start ** i) for i in range(n)]
+
+    if math.log2(num_heads).is_integer():
+        s = pow2_slopes(num_heads)
+    else:
+        p = 2 ** math.floor(math.log2(num_heads))
+        extra = pow2_slopes(2 * p)[0::2][: num_heads - p]
+        s = pow2_slopes(p) + extra
+    return torch.tensor(s, dtype=torch.float32)
+
+
+def attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    slopes: Optional[torch.Tensor] = None,
+    dropout_p: float = 0.0,
+    training: bool = False,
+) -> torch.Tensor:
+    """Causal multi-head attention with optional ALiBi bias, fp32 softmax.
+
+    q, k, v: (B, H, T, D). Returns (B, H, T, D) in q.dtype.
+
+    ALiBi bias for query i, key j (j <= i): slope_h * (j - i)  (<= 0).
+    The reference applies the shift-invariant single-row form
+    (layers.py:33-44); per-row they differ by a constant, so softmax output
+    is identical.
+    """
+    B, H, T, D = q.shape
+    scores = (q.float() @ k.float().transpose(-1, -2)) / math.sqrt(D)
+    if slopes is not None:
+        pos = torch.arange(T, device=q.device, dtype=torch.float32)
+        rel = pos.view(1, T) - pos.view(T, 1)  # (j - i), negative below diag
+        scores = scores + slopes.to(q.device).float().view(1, H, 1, 1) * rel.view(1, 1, T, T)
+    causal = torch.ones(T, T, dtype=torch.bool, device=q.device).tril()
+    scores = scores.masked_fill(~causal.view(1, 1, T, T), torch.finfo(torch.float32).min)
+    probs = F.softmax(scores, dim=-1)
+    if dropout_p > 0.0 and training:
+        probs = F.dropout(probs, p=dropout_p)
+    out = probs.to(v.dtype) @ v
+    return out.to(q.dtype)
+
+
+def layer_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    """Bias-free LayerNorm in fp32, cast back to input dtype.
+
+    Matches flax nn.LayerNorm(use_bias=False) (reference src/models/GPT.py:42).
+    """
+    y = F.layer_norm(x.float(), (x.shape[-1],), weight=weight.float(), bias=None, eps=eps)
+    return y.to(x.dtype)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    """tanh-approximate GELU (flax default, reference layers.py:68)."""
+    return F.gelu(x, approximate="tanh")
+
+
+def cross_entropy(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+    """Mean cross entropy over all positions, fp32 log-softmax.
+
+    logits: (N, V), targets: (N,) int64. Equivalent to the reference's
+    one-hot * log_softmax mean (src/utils/losses.py:10-23) without
+    materializing the one-hot.
+    """
+    return F.cross_entropy(logits.float(), targets)
+
+
+def adamw_update(
+    param_f32: torch.Tensor,
+    grad: torch.Tensor,
+    exp_avg: torch.Tensor,
+    exp_avg_sq: torch.Tensor,
+    step: int,
+    lr: float,
+    beta1: float,
+    beta2: float,
+    eps: float,
+    weight_decay: float,
+    clip_value: float = 1.0,
+    grad_scale: float = 1.0,
+) -> None:
+    """In-place AdamW on an fp32 master shard, reference semantics.
+
+    Order matches the reference optax chain (main_zero.py:160-168):
+    grad * grad_scale (the 1/accum divide of xmap_train_functions.py:81),
+    element-wise clip of the gradient to +-clip_value (optax.clip(1.0) —
+    NOT global-norm clipping), then scale_by_adam(b2=0.95), then masked
+    weight decay, then -lr scaling.  Decoupled decay: update includes
+    weight_decay * param (optax.add_decayed_weights semantics).
+    """
+    g = grad.float() * grad_scale
+    if clip_value:
+        g.clamp_(-clip_value, clip_value)
+    exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
+    exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    bc1 = 1 - beta1 ** step
+    bc2 = 1 - beta2 ** step
+    m_hat = exp_avg / bc1
+    v_hat = exp_avg_sq / bc2
+    update = m_hat / (v_hat.sqrt() + eps)
+    if weight_decay:
+        update = update + weight_decay * param_f32
+    param_f32.add_(update, alpha=-lr)

@@ -1,0 +1,96 @@
+"""Backend-agnostic collectives for the ZeRO-1 engine.
+
+On GPU this is RCCL over xGMI (torch.distributed backend "nccl" IS RCCL on
+ROCm); on CPU (plumbing tests, BASELINE config #1) it is gloo, which lacks
+reduce_scatter_tensor / all_gather_into_tensor — those fall back to
+all_reduce + slice / all_gather-list, functionally identical.
+
+Replaces the reference's XLA collectives: the grad pmean +
+pjit reshard (xmap_train_functions.py:84, main_zero.py:458-460) become ONE
+reduce-scatter; the replicated out_shardings all-gather (main_zero.py:455)
+becomes an explicit all-gather of updated params.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+
+def is_dist() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def world_size() -> int:
+    return dist.get_world_size() if is_dist() else 1
+
+
+def rank() -> int:
+    return dist.get_rank() if is_dist() else 0
+
+
+def backend() -> str:
+    return dist.get_backend() if is_dist() else "none"
+
+
+def _supports_native_rs() -> bool:
+    return is_dist() and dist.get_backend() != "gloo"
+
+
+def reduce_scatter_mean(
+    out_shard: torch.Tensor, flat: torch.Tensor, async_op: bool = False
+):
+    """out_shard <- mean over ranks of this rank's slice of `flat`.
+
+    flat: (world * shard_numel,) contiguous; out_shard: (shard_numel,).
+    Returns a work handle when async_op (nccl path) else None.
+    """
+    ws = world_size()
+    if ws == 1:
+        out_shard.copy_(flat)
+        return None
+    if _supports_native_rs():
+        return dist.reduce_scatter_tensor(
+            out_shard, flat, op=dist.ReduceOp.AVG, async_op=async_op
+        )
+    # gloo fallback: all_reduce then local slice
+    dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+    r = rank()
+    n = out_shard.numel()
+    out_shard.copy_(flat[r * n : (r + 1) * n])
+    out_shard.div_(ws)
+    return None
+
+
+def all_gather_flat(flat: torch.Tensor, shard: torch.Tensor, async_op: bool = False):
+    """flat <- concat over ranks of `shard` (inverse of the scatter)."""
+    ws = world_size()
+    if ws == 1:
+        flat.copy_(shard)
+        return None
+    if _supports_native_rs():
+        return dist.all_gather_into_tensor(flat, shard.contiguous(), async_op=async_op)
+    chunks = list(flat.chunk(ws))
+    dist.all_gather(chunks, shard.contiguous())
+    return None
+
+
+def all_reduce_mean_(t: torch.Tensor) -> torch.Tensor:
+    """In-place mean all-reduce (the loss pmean, xmap_train_functions.py:83)."""
+    if world_size() > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        t.div_(world_size())
+    return t
+
+
+def broadcast_(t: torch.Tensor, src: int = 0) -> torch.Tensor:
+    if world_size() > 1:
+        dist.broadcast(t, src=src)
+    return t
+
+
+def barrier() -> None:
+    if world_size() > 1:
+        dist.barrier()

@@ -1,0 +1,283 @@
+"""ZeRO stage-1 optimizer: bucketed reduce-scatter + sharded AdamW + all-gather.
+
+MI355X-native redesign of the reference's xmap/pjit ZeRO-1
+(src/partitioning/partition.py:49-140, xmap_train_functions.py:110-123,
+main_zero.py:438-460). Instead of regex PartitionSpecs + implicit XLA
+resharding, parameters are flattened into contiguous buckets; each rank owns
+the [rank * n/ws, (rank+1) * n/ws) slice of every bucket:
+
+  backward        -> grads accumulate into flat bf16 bucket views
+                     (comm deferred to the last micro-step, matching the
+                     reference's accumulate-then-pmean, xmap:74-84)
+  last micro-step -> per-bucket RCCL reduce-scatter(AVG) on a side HIP
+                     stream, overlapped with the rest of backward
+  step()          -> fused AdamW HIP kernel on each fp32 master shard
+                     (element-wise clip(1.0), b2=0.95, masked weight decay),
+                     emitting the updated bf16 working shard
+                  -> RCCL all-gather of updated bf16 shards back into the
+                     flat param buffers (the out_shardings=None all-gather
+                     of main_zero.py:455), overlapped bucket-by-bucket
+
+Buckets are split by weight-decay group (decay = ndim > 1, the reference's
+mask at main_zero.py:154-158; no (block,embed) positional embedding exists
+here since positioning is ALiBi), so the fused kernel needs no per-element
+decay mask. Bucket size defaults to ~100 MB — sized so each of the 7 xGMI
+p2p links carries a parallel chunk of a ring reduce-scatter.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional, Sequence, Tuple
+
+import torch
+
+from .. import ops
+from . import comm
+
+ALIGN = 128  # bucket padding granule (elements)
+
+
+@dataclass
+class Bucket:
+    idx: int
+    decay: bool
+    names: List[str]
+    params: List[torch.nn.Parameter]
+    offsets: List[int]  # start offset of each param in the flat buffer
+    numel: int  # padded
+    flat_param: torch.Tensor = None  # working dtype, full bucket (replicated)
+    flat_grad: torch.Tensor = None  # working dtype, full bucket
+    grad_shard: torch.Tensor = None  # working dtype, shard
+    master: torch.Tensor = None  # fp32 shard
+    exp_avg: torch.Tensor = None  # fp32 shard
+    exp_avg_sq: torch.Tensor = None  # fp32 shard
+    ready: int = 0
+    rs_work: object = None
+
+
+class ZeRO1Optimizer:
+    """ZeRO-1 sharded AdamW over flat bucketed parameters."""
+
+    def __init__(
+        self,
+        named_params: Sequence[Tuple[str, torch.nn.Parameter]],
+        lr: Callable[[int], float] | float,
+        betas: Tuple[float, float] = (0.9, 0.95),
+        eps: float = 1e-8,
+        weight_decay: float = 0.1,
+        clip_value: float = 1.0,
+        bucket_mb: float = 100.0,
+        param_dtype: Optional[torch.dtype] = None,
+        accum_steps: int = 1,
+        overlap_comm: bool = True,
+    ):
+        named_params = [(n, p) for n, p in named_params if p.requires_grad]
+        # dedup shared (tied) parameters, keeping first name
+        seen, uniq = set(), []
+        for n, p in named_params:
+            if id(p) not in seen:
+                seen.add(id(p))
+                uniq.append((n, p))
+        if not uniq:
+            raise ValueError("no trainable parameters")
+        self.lr = lr if callable(lr) else (lambda _s, _v=lr: _v)
+        self.beta1, self.beta2 = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.clip_value = clip_value
+        self.accum_steps = accum_steps
+        self.step_count = 0
+        self.world = comm.world_size()
+        self.rank = comm.rank()
+        self.device = uniq[0][1].device
+        self.param_dtype = param_dtype or uniq[0][1].dtype
+        self.overlap_comm = overlap_comm and self.device.type == "cuda" and self.world > 1
+        self._sync = accum_steps == 1
+        self._comm_stream = (
+            torch.cuda.Stream(device=self.device) if self.overlap_comm else None
+        )
+
+        self.buckets: List[Bucket] = self._build_buckets(uniq, int(bucket_mb * 1e6))
+        self._param_bucket: Dict[int, Tuple[Bucket, int]] = {}
+        for b in self.buckets:
+            for i, p in enumerate(b.params):
+                self._param_bucket[id(p)] = (b, i)
+        self._install_views_and_hooks()
+
+    # ------------------------------------------------------------------
+    def _build_buckets(self, named, bucket_bytes: int) -> List[Bucket]:
+        esize = torch.tensor([], dtype=self.param_dtype).element_size()
+        cap = max(bucket_bytes // esize, ALIGN)
+        groups = {True: [], False: []}
+        for n, p in named:
+            groups[p.dim() > 1].append((n, p))
+        buckets: List[Bucket] = []
+        for decay, items in groups.items():
+            cur: List[Tuple[str, torch.nn.Parameter]] = []
+            cur_n = 0
+            for n, p in items:
+                if cur and cur_n + p.numel() > cap:
+                    buckets.append(self._make_bucket(len(buckets), decay, cur))
+                    cur, cur_n = [], 0
+                cur.append((n, p))
+                cur_n += p.numel()
+            if cur:
+                buckets.append(self._make_bucket(len(buckets), decay, cur))
+        return buckets
+
+    def _make_bucket(self, idx: int, decay: bool, items) -> Bucket:
+        names = [n for n, _ in items]
+        params = [p for _, p in items]
+        offsets, off = [], 0
+        for p in params:
+            offsets.append(off)
+            off += p.numel()
+        pad_to = ALIGN * self.world
+        numel = (off + pad_to - 1) // pad_to * pad_to
+        b = Bucket(idx=idx, decay=decay, names=names, params=params, offsets=offsets, numel=numel)
+        dev, wd = self.device, self.param_dtype
+        shard_n = numel // self.world
+        b.flat_param = torch.zeros(numel, dtype=wd, device=dev)
+        b.flat_grad = torch.zeros(numel, dtype=wd, device=dev)
+        b.grad_shard = torch.zeros(shard_n, dtype=wd, device=dev)
+        # fp32 master shard initialized from the (possibly fp32) params
+        master_full = torch.zeros(numel, dtype=torch.float32, device=dev)
+        for p, o in zip(params, offsets):
+            master_full[o : o + p.numel()].copy_(p.data.reshape(-1).float())
+        b.master = master_full[self.rank * shard_n : (self.rank + 1) * shard_n].clone()
+        b.flat_param.copy_(master_full.to(wd))
+        del master_full
+        b.exp_avg = torch.zeros(shard_n, dtype=torch.float32, device=dev)
+        b.exp_avg_sq = torch.zeros(shard_n, dtype=torch.float32, device=dev)
+        return b
+
+    def _install_views_and_hooks(self):
+        for b in self.buckets:
+            for p, o in zip(b.params, b.offsets):
+                shape = p.data.shape
+                p.data = b.flat_param[o : o + p.numel()].view(shape)
+                p.grad = b.flat_grad[o : o + p.numel()].view(shape)
+                p.register_post_accumulate_grad_hook(self._on_grad_ready)
+
+    # ------------------------------------------------------------------
+    def set_sync(self, sync: bool):
+        """Enable/disable grad communication (False on non-final micro-steps)."""
+        self._sync = sync
+
+    def _on_grad_ready(self, p: torch.nn.Parameter):
+        if not self._sync:
+            return
+        b, _ = self._param_bucket[id(p)]
+        b.ready += 1
+        if b.ready == len(b.params):
+            self._launch_reduce(b)
+
+    def _launch_reduce(self, b: Bucket):
+        if self.overlap_comm:
+            self._comm_stream.wait_stream(torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(self._comm_stream):
+                b.rs_work = comm.reduce_scatter_mean(b.grad_shard, b.flat_grad, async_op=False)
+        else:
+            b.rs_work = comm.reduce_scatter_mean(b.grad_shard, b.flat_grad, async_op=False)
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def step(self, closure=None):
+        assert self._sync, "step() called while grad sync disabled"
+        self.step_count += 1
+        lr = float(self.lr(self.step_count))
+        # any bucket whose hook never fired (e.g. unused param) — reduce now
+        for b in self.buckets:
+            if b.ready != len(b.params):
+                self._launch_reduce(b)
+        if self.overlap_comm:
+            torch.cuda.current_stream(self.device).wait_stream(self._comm_stream)
+        shard_n = None
+        for b in self.buckets:
+            shard_n = b.numel // self.world
+            own = b.flat_param[self.rank * shard_n : (self.rank + 1) * shard_n]
+            ops.adamw_step(
+                b.master,
+                own if self.param_dtype == torch.bfloat16 else None,
+                b.grad_shard,
+                b.exp_avg,
+                b.exp_avg_sq,
+                self.step_count,
+                lr,
+                self.beta1,
+                self.beta2,
+                self.eps,
+                self.weight_decay if b.decay else 0.0,
+                self.clip_value,
+                grad_scale=1.0 / self.accum_steps,
+            )
+            if self.param_dtype != torch.bfloat16:
+                own.copy_(b.master.to(self.param_dtype))
+            comm.all_gather_flat(b.flat_param, own)
+        self.zero_grad()
+        return lr
+
+    @torch.no_grad()
+    def zero_grad(self, set_to_none: bool = False):
+        for b in self.buckets:
+            b.flat_grad.zero_()
+            b.ready = 0
+            b.rs_work = None
+
+    # ------------------------------------------------------------------
+    # Checkpointing (reference main_zero.py:58-139 two-stream save/restore)
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def full_param_state_dict(self) -> Dict[str, torch.Tensor]:
+        """Gather fp32 master params to a full state_dict (rank 0 usable)."""
+        out: Dict[str, torch.Tensor] = {}
+        for b in self.buckets:
+            full = torch.zeros(b.numel, dtype=torch.float32, device=self.device)
+            comm.all_gather_flat(full, b.master)
+            for n, p, o in zip(b.names, b.params, b.offsets):
+                out[n] = full[o : o + p.numel()].view(p.shape).cpu().clone()
+            del full
+        return out
+
+    @torch.no_grad()
+    def optimizer_state_dict(self) -> Dict:
+        """Gather full optimizer state (Adam moments) to CPU (rank 0 usable)."""
+        state = {"step": self.step_count, "buckets": []}
+        for b in self.buckets:
+            mu = torch.zeros(b.numel, dtype=torch.float32, device=self.device)
+            nu = torch.zeros(b.numel, dtype=torch.float32, device=self.device)
+            comm.all_gather_flat(mu, b.exp_avg)
+            comm.all_gather_flat(nu, b.exp_avg_sq)
+            state["buckets"].append(
+                {"names": b.names, "exp_avg": mu.cpu(), "exp_avg_sq": nu.cpu()}
+            )
+            del mu, nu
+        return state
+
+    @torch.no_grad()
+    def load_param_state_dict(self, sd: Dict[str, torch.Tensor]):
+        """Load full fp32 params: refill master shards + bf16 working copy."""
+        shard = lambda t, n: t[self.rank * (n // self.world) : (self.rank + 1) * (n // self.world)]
+        for b in self.buckets:
+            full = torch.zeros(b.numel, dtype=torch.float32, device=self.device)
+            for n, p, o in zip(b.names, b.params, b.offsets):
+                full[o : o + p.numel()].copy_(sd[n].reshape(-1).float().to(self.device))
+            b.master.copy_(shard(full, b.numel))
+            b.flat_param.copy_(full.to(self.param_dtype))
+            del full
+
+    @torch.no_grad()
+    def load_optimizer_state_dict(self, state: Dict):
+        self.step_count = state["step"]
+        shard = lambda t, n: t[self.rank * (n // self.world) : (self.rank + 1) * (n // self.world)]
+        for b, bs in zip(self.buckets, state["buckets"]):
+            assert b.names == bs["names"], "bucket layout mismatch on restore"
+            b.exp_avg.copy_(shard(bs["exp_avg"].to(self.device), b.numel))
+            b.exp_avg_sq.copy_(shard(bs["exp_avg_sq"].to(self.device), b.numel))
+
+    # convenience for logging / tests
+    @property
+    def num_shard_elems(self) -> int:
+        return sum(b.numel // self.world for b in self.buckets)

@@ -1,0 +1,86 @@
+"""Training engine: grad-accumulated ZeRO-1 train/eval steps.
+
+MI355X-native replacement for the reference's xmap/pjit step functions
+(src/partitioning/xmap_train_functions.py:26-123): the fori_loop becomes a
+Python micro-batch loop with grad communication deferred to the last
+micro-step (ZeRO1Optimizer.set_sync); pmean(loss) is an explicit RCCL
+all-reduce; the sharded AdamW + param all-gather live in ZeRO1Optimizer.step.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict
+
+import numpy as np
+import torch
+
+from ..parallel import comm
+from ..parallel.zero import ZeRO1Optimizer
+
+
+def reshape_context(tokens: torch.Tensor, train_context: int) -> torch.Tensor:
+    """Seq-len curriculum: (B, max_ctx) -> (B * max_ctx/train_ctx, train_ctx)
+    when train_context < max_ctx (reference main_zero.py:477-478)."""
+    B, ctx = tokens.shape
+    if train_context < ctx:
+        assert ctx % train_context == 0
+        return tokens.reshape(B * (ctx // train_context), train_context)
+    return tokens
+
+
+class TrainEngine:
+    def __init__(
+        self,
+        model: torch.nn.Module,
+        optimizer: ZeRO1Optimizer,
+        accum_steps: int,
+        train_context: int,
+        device: torch.device,
+    ):
+        self.model = model
+        self.optimizer = optimizer
+        self.accum_steps = accum_steps
+        self.train_context = train_context
+        self.device = device
+
+    def _to_device(self, batch) -> torch.Tensor:
+        if isinstance(batch, np.ndarray):
+            batch = torch.from_numpy(batch)
+        return batch.to(self.device, dtype=torch.long, non_blocking=True)
+
+    def train_step(self, batch) -> Dict[str, float]:
+        """One optimizer step over `batch` ((B, max_ctx) tokens)."""
+        self.model.train()
+        tokens = reshape_context(self._to_device(batch), self.train_context)
+        rows = tokens.shape[0]
+        assert rows % self.accum_steps == 0, (
+            f"batch rows {rows} not divisible by accum {self.accum_steps}"
+        )
+        micros = tokens.chunk(self.accum_steps)
+        loss_sum = torch.zeros((), device=self.device, dtype=torch.float32)
+        for i, mb in enumerate(micros):
+            self.optimizer.set_sync(i == self.accum_steps - 1)
+            _, loss = self.model(mb, labels=mb)
+            loss.backward()
+            loss_sum += loss.detach().float()
+        lr = self.optimizer.step()
+        loss_mean = loss_sum / self.accum_steps
+        comm.all_reduce_mean_(loss_mean)  # pmean(loss), xmap:83
+        loss_val = float(loss_mean.item())
+        return {
+            "train/loss": loss_val,
+            "train/ppl": math.exp(min(loss_val, 30.0)),
+            "lr": lr,
+        }
+
+    @torch.no_grad()
+    def eval_step(self, batch) -> Dict[str, float]:
+        """Loss on a validation batch (reference eval_step, xmap:94-107)."""
+        self.model.eval()
+        tokens = reshape_context(self._to_device(batch), self.train_context)
+        _, loss = self.model(tokens, labels=tokens)
+        loss = loss.detach().float()
+        comm.all_reduce_mean_(loss)
+        v = float(loss.item())
+        return {"validation/loss": v, "validation/ppl": math.exp(min(v, 30.0))}

@@ -1,0 +1,51 @@
+"""YAML config loading with attribute access.
+
+Keeps the reference's config surface (conf/config.yaml + conf/model_config.yaml,
+see reference main_zero.py:41-55 and src/models/GPT.py:116-137) without the
+OmegaConf dependency: plain pyyaml into a dot-accessible dict.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict
+
+import yaml
+
+
+class DotDict(dict):
+    """dict with attribute access, recursively wrapping nested dicts."""
+
+    def __getattr__(self, name: str) -> Any:
+        try:
+            return self[name]
+        except KeyError as e:
+            raise AttributeError(name) from e
+
+    def __setattr__(self, name: str, value: Any) -> None:
+        self[name] = value
+
+    @classmethod
+    def wrap(cls, obj: Any) -> Any:
+        if isinstance(obj, dict):
+            return cls({k: cls.wrap(v) for k, v in obj.items()})
+        if isinstance(obj, (list, tuple)):
+            return type(obj)(cls.wrap(v) for v in obj)
+        return obj
+
+
+def load_config(path: str) -> DotDict:
+    with open(path, "r") as f:
+        raw = yaml.safe_load(f)
+    return DotDict.wrap(raw or {})
+
+
+def flatten_dict(d: Dict[str, Any], parent_key: str = "", sep: str = ".") -> Dict[str, Any]:
+    """Dotted-key flatten of a nested config (reference src/utils/configs.py:7-17)."""
+    items = []
+    for k, v in d.items():
+        key = f"{parent_key}{sep}{k}" if parent_key else str(k)
+        if isinstance(v, dict):
+            items.extend(flatten_dict(v, key, sep=sep).items())
+        else:
+            items.append((key, v))
+    return dict(items)

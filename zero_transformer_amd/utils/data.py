@@ -1,0 +1,148 @@
+"""Data pipeline: rank-sharded token streams.
+
+Replaces the reference's WebDataset-from-GCS pipeline (main_zero.py:377-421)
+with two MI355X-box-friendly sources:
+
+  * SyntheticTokens — random token sequences of the packed shape, for
+    benchmarking and tests (no network on the GPU boxes).
+  * IndexedTarTokens — WebDataset-style local .tar shards of .npy token
+    arrays listed in a newline index file (the reference's data/index/*.index
+    role, reference main_zero.py:189-198), sharded across ranks
+    (split_by_jax_process equivalent, main_zero.py:377-387) and across
+    DataLoader workers.
+
+Both yield int32/int64 token arrays of length max_context; the training
+driver applies the seq-len curriculum reshape (2048 -> train_context rows,
+reference main_zero.py:477-493).
+"""
+
+from __future__ import annotations
+
+import io
+import os
+import tarfile
+from typing import Iterator, List, Optional
+
+import numpy as np
+import torch
+from torch.utils.data import DataLoader, IterableDataset
+
+
+def numpy_collate(batch):
+    """numpy-stacking collate (reference src/utils/dataloader.py:9-16)."""
+    if isinstance(batch[0], np.ndarray):
+        return np.stack(batch)
+    if isinstance(batch[0], (list, tuple)):
+        return type(batch[0])(numpy_collate(x) for x in zip(*batch))
+    return np.asarray(batch)
+
+
+class SyntheticTokens(IterableDataset):
+    """Endless random token stream with a fixed per-epoch length."""
+
+    def __init__(self, vocab_size: int, max_context: int, samples: int, seed: int = 0):
+        self.vocab_size = vocab_size
+        self.max_context = max_context
+        self.samples = samples
+        self.seed = seed
+
+    def __iter__(self) -> Iterator[np.ndarray]:
+        info = torch.utils.data.get_worker_info()
+        wid = info.id if info else 0
+        rng = np.random.default_rng(self.seed + 1000003 * wid)
+        for _ in range(self.samples):
+            yield rng.integers(
+                0, self.vocab_size, size=(self.max_context,), dtype=np.int64
+            )
+
+    def __len__(self):
+        return self.samples
+
+
+class IndexedTarTokens(IterableDataset):
+    """Stream .npy token arrays out of local tar shards listed in an index.
+
+    Index file: one shard path per line (the reference's gs:// URL lists,
+    data/index/*.index — here local filesystem paths). Shards are dealt
+    round-robin to (rank, worker) pairs; decode errors are skipped
+    (wds.warn_and_continue equivalent, reference main_zero.py:392).
+    """
+
+    def __init__(
+        self,
+        index_path: str,
+        max_context: int,
+        rank: int = 0,
+        world_size: int = 1,
+        seed: int = 0,
+        shuffle: bool = True,
+    ):
+        with open(index_path) as f:
+            self.shards = [ln.strip() for ln in f if ln.strip()]
+        if not self.shards:
+            raise ValueError(f"empty index: {index_path}")
+        self.max_context = max_context
+        self.rank = rank
+        self.world_size = world_size
+        self.seed = seed
+        self.shuffle = shuffle
+
+    def _iter_shard(self, path: str) -> Iterator[np.ndarray]:
+        try:
+            with tarfile.open(path, "r") as tf:
+                for member in tf:
+                    if not member.isfile():
+                        continue
+                    try:
+                        buf = tf.extractfile(member).read()
+                        arr = np.load(io.BytesIO(buf), allow_pickle=False)
+                        arr = np.asarray(arr).reshape(-1).astype(np.int64)
+                        for s in range(0, arr.size - self.max_context + 1, self.max_context):
+                            yield arr[s : s + self.max_context]
+                    except Exception:
+                        continue  # warn_and_continue semantics
+        except Exception:
+            return
+
+    def __iter__(self) -> Iterator[np.ndarray]:
+        info = torch.utils.data.get_worker_info()
+        nworkers = info.num_workers if info else 1
+        wid = info.id if info else 0
+        stride = self.world_size * nworkers
+        offset = self.rank * nworkers + wid
+        shards = list(self.shards)
+        if self.shuffle:
+            rng = np.random.default_rng(self.seed)
+            rng.shuffle(shards)
+        for shard in shards[offset::stride]:
+            yield from self._iter_shard(shard)
+
+
+def make_loader(
+    dataset: IterableDataset,
+    batch_size: int,
+    num_workers: int = 2,
+) -> DataLoader:
+    return DataLoader(
+        dataset,
+        batch_size=batch_size,
+        collate_fn=numpy_collate,
+        num_workers=num_workers,
+        drop_last=True,
+        persistent_workers=num_workers > 0,
+    )
+
+
+def build_dataset(cfg, split: str, rank: int, world_size: int, model_cfg=None):
+    """Pick the data source from cfg.data (reference main_zero.py:377-421)."""
+    data = cfg.data
+    if data.corpus == "synthetic":
+        vocab = model_cfg.vocab_size if model_cfg is not None else 50304
+        samples = int(data.train_samples) // max(world_size, 1)
+        if split != "train":
+            samples = min(samples, 4096)
+        return SyntheticTokens(vocab, int(data.max_context), samples, seed=0 if split == "train" else 1)
+    index = data.index_path_train if split == "train" else data.index_path_validation
+    return IndexedTarTokens(
+        index, int(data.max_context), rank=rank, world_size=world_size
+    )
