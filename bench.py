@@ -1,0 +1,130 @@
+"""Flagship training-step benchmark (driver contract).
+
+    python bench.py --gpus N --steps K --warmup W
+
+Measures the whole-job training throughput (tokens/s over all N GPUs) of the
+headline config: GPT-1.3B, ZeRO-1, bf16, ALiBi, block_size 2048, synthetic
+data, random-init weights (BASELINE.json). Weak scaling: per-GPU batch is
+fixed as N grows. For N > 1 the driver launches this script under
+torch.distributed.run with one rank per GPU over RCCL.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+BASELINE_TOKS = 131000.0  # reference TPU v3-32 pod-level tokens/s (BASELINE.md, 760M derived)
+
+
+def parse():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", default="1_3b_2048")
+    p.add_argument("--batch", type=int, default=16, help="per-GPU sequences per step")
+    p.add_argument("--accum", type=int, default=1)
+    p.add_argument("--bucket-mb", type=float, default=100.0)
+    return p.parse_args()
+
+
+def main():
+    args = parse()
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    assert torch.cuda.is_available(), "bench.py requires MI355X GPUs"
+    if world > 1:
+        dist.init_process_group("nccl", rank=rank, world_size=world)
+    device = torch.device("cuda", local_rank)
+    torch.cuda.set_device(device)
+
+    from zero_transformer_amd.models import model_getter
+    from zero_transformer_amd.parallel.zero import ZeRO1Optimizer
+    from zero_transformer_amd.training.trainer import TrainEngine
+    from zero_transformer_amd.utils.lr import warmup_cosine
+
+    torch.manual_seed(1234 + rank)
+    model, mcfg = model_getter(args.model, return_cfg=True)
+    model = model.to(device)
+    n_params = model.num_params()
+    seq = int(mcfg.block_size)
+    opt = ZeRO1Optimizer(
+        list(model.named_parameters()),
+        lr=warmup_cosine(3e-4, 2000, 143000, 3e-5),
+        betas=(0.9, 0.95),
+        weight_decay=0.1,
+        clip_value=1.0,
+        bucket_mb=args.bucket_mb,
+        param_dtype=torch.bfloat16,
+        accum_steps=args.accum,
+    )
+    engine = TrainEngine(model, opt, args.accum, seq, device)
+
+    # synthetic data, device-resident (BASELINE: synthetic / random-init)
+    gen = torch.Generator(device="cpu").manual_seed(99 + rank)
+    batches = [
+        torch.randint(0, mcfg.vocab_size, (args.batch, seq), generator=gen).to(device)
+        for _ in range(2)
+    ]
+
+    for i in range(args.warmup):
+        engine.train_step(batches[i % 2])
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        engine.train_step(batches[i % 2])
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    tokens_per_step = args.batch * seq * world  # whole-job
+    value = tokens_per_step / (elapsed / args.steps)
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "training throughput (tokens/s, whole job)",
+                    "value": round(value, 1),
+                    "unit": "tokens/s",
+                    "n_gpus": world,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": round(ms_per_step, 2),
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": round(value / BASELINE_TOKS, 3),
+                    "dtype": "bf16",
+                    "data": "synthetic",
+                    "config": {
+                        "model": f"GPT-1.3B ({n_params/1e9:.2f}B params, {args.model})",
+                        "global_batch": args.batch * world * args.accum,
+                        "seq_len": seq,
+                        "parallelism": f"dp{world}+zero1",
+                        "grad_accum": args.accum,
+                        "dropout": float(mcfg.dropout),
+                    },
+                }
+            )
+        )
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,92 @@
+"""Build the in-tree HIP extension (_zta_hip) for gfx950 with hipcc.
+
+No hipify, no CUDA shims: sources are native HIP/CDNA4 (.hip) compiled
+directly by hipcc and linked against libtorch. The resulting .so lives
+in-tree (zero_transformer_amd/ops/) so it travels with repo snapshots.
+
+Usage:  python -m zero_transformer_amd.ops.build [--force]
+"""
+
+from __future__ import annotations
+
+import hashlib
+import os
+import subprocess
+import sys
+import sysconfig
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(HERE, "csrc")
+OUT = os.path.join(HERE, "_zta_hip.so")
+STAMP = os.path.join(HERE, ".build_stamp")
+
+SOURCES = [
+    "bindings.cpp",
+    "layernorm.hip",
+    "gelu.hip",
+    "cross_entropy.hip",
+    "adamw.hip",
+    "attention_fwd.hip",
+    "attention_bwd.hip",
+]
+
+
+def _torch_flags():
+    import torch
+    import torch.utils.cpp_extension as ce
+
+    includes = ce.include_paths() + [sysconfig.get_paths()["include"]]
+    libdirs = ce.library_paths()
+    abi = int(torch._C._GLIBCXX_USE_CXX11_ABI)
+    return includes, libdirs, abi
+
+
+def _src_hash() -> str:
+    h = hashlib.sha256()
+    for s in SOURCES + ["common.h"]:
+        p = os.path.join(CSRC, s)
+        if os.path.exists(p):
+            h.update(open(p, "rb").read())
+    return h.hexdigest()
+
+
+def build(force: bool = False, verbose: bool = True) -> str:
+    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+    want = _src_hash()
+    if not force and os.path.exists(OUT) and os.path.exists(STAMP):
+        if open(STAMP).read().strip() == want:
+            return OUT
+    includes, libdirs, abi = _torch_flags()
+    hipcc = os.environ.get("HIPCC", "hipcc")
+    cmd = [
+        hipcc,
+        "--offload-arch=gfx950",
+        "-O3",
+        "-std=c++17",
+        "-fPIC",
+        "-shared",
+        f"-D_GLIBCXX_USE_CXX11_ABI={abi}",
+        "-DTORCH_EXTENSION_NAME=_zta_hip",
+        "-DTORCH_API_INCLUDE_EXTENSION_H",
+        "-DUSE_ROCM",
+        "-fno-gpu-rdc",
+        "-Wno-unused-result",
+    ]
+    for i in includes:
+        cmd += ["-I", i]
+    cmd += [os.path.join(CSRC, s) for s in SOURCES]
+    for d in libdirs:
+        cmd += ["-L", d, f"-Wl,-rpath,{d}"]
+    cmd += ["-ltorch", "-ltorch_python", "-ltorch_hip", "-lc10", "-lc10_hip", "-lamdhip64"]
+    cmd += ["-o", OUT]
+    if verbose:
+        print("[zta build]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    with open(STAMP, "w") as f:
+        f.write(want)
+    return OUT
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)
+    print(OUT)

@@ -1,0 +1,219 @@
+// Fused causal ALiBi flash attention forward for gfx950 (CDNA4 MFMA).
+//
+// Replaces the reference's dense O(T^2) materialized attention
+// (layers.py:159-178: q@k^T, +alibi, tril mask, fp32 softmax, dropout, @v)
+// with a flash-style online-softmax kernel: per 128-row Q block, iterate
+// 32-key K/V tiles with running (m, l) in fp32 — the softmax statistics stay
+// fp32 end to end (the bf16-softmax failure of logs/580.md:94-98 cannot
+// happen by construction). ALiBi is applied as the true bias
+// slope*(j - i) (shift-equivalent to the reference's single-row mask,
+// layers.py:33-44). Dropout uses a counter-based RNG so backward
+// regenerates the identical mask without storing it.
+//
+// Structure: 4 waves/block, each wave owns 32 q rows via one
+// v_mfma_f32_32x32x16_bf16 accumulator; K staged row-major in XOR-swizzled
+// LDS (T2: conflict-free ds_read_b128 B-fragments), V staged transposed
+// ([d][key]) so the PV B-operand reads are k-contiguous.
+//
+// Fragment maps (gfx950, §3 of the CDNA4 guide):
+//   A: lane l holds A[i = l%32][k = 8*(l/32) + e], e = 0..7
+//   B: lane l holds B[k = 8*(l/32) + e][j = l%32]
+//   C/D: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+
+#include "common.h"
+
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+constexpr int QB = 128;  // q rows per block (4 waves x 32)
+constexpr int KB = 32;   // kv tile
+constexpr float NEG_INF = -3.0e38f;
+
+// XOR swizzle for 256-byte-stride LDS rows (T2): spreads ds_read_b128 lane
+// groups over 8 slots. Applied identically on write and read.
+ZTA_DEV int swz(int row, int byte_off) { return byte_off ^ ((row & 7) << 4); }
+
+// K tile: [KB][128] bf16, row stride 256 B, swizzled.
+// Vt tile: [D][KB + 8] bf16, row stride 80 B (bank-conflict-free without swizzle).
+// P tile (per wave): [32][KB + 8] bf16.
+constexpr int VT_STRIDE = KB + 8;  // elements
+
+template <int D>
+__global__ __launch_bounds__(256) void flash_fwd_kernel(
+    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
+    const uint16_t* __restrict__ v, const float* __restrict__ slopes,
+    uint16_t* __restrict__ o, float* __restrict__ lse, int H, int T,
+    float scale, float p_drop, uint64_t seed) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint16_t* k_lds = (uint16_t*)smem;                     // KB*128
+  uint16_t* vt_lds = k_lds + KB * 128;                   // D*VT_STRIDE
+  uint16_t* p_lds = vt_lds + D * VT_STRIDE;              // 4*32*VT_STRIDE
+
+  const int bh = blockIdx.y;
+  const int h = bh % H;
+  const long base = (long)bh * T * D;
+  const int q0 = blockIdx.x * QB;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int li = lane & 31;   // A-operand row / C col
+  const int hi = lane >> 5;   // half-wave
+  const int qw = q0 + wave * 32;  // this wave's first q row
+  const float slope = slopes[h];
+  const float inv_keep = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
+
+  // ---- Q fragments in registers: A[i=li][kk = s*16 + 8*hi + e] ----
+  constexpr int KS = D / 16;  // QK^T k-steps
+  bf16x8 q_frag[KS];
+  {
+    const int qi = qw + li;
+    const bool ok = qi < T;
+#pragma unroll
+    for (int s = 0; s < KS; ++s) {
+      if (ok) {
+        q_frag[s] = *reinterpret_cast<const bf16x8*>(&q[base + (long)qi * D + s * 16 + 8 * hi]);
+      } else {
+        q_frag[s] = bf16x8{};
+      }
+    }
+  }
+
+  float m_run[16], l_run[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    m_run[r] = NEG_INF;
+    l_run[r] = 0.f;
+  }
+  constexpr int DB = D / 32;  // PV output column blocks
+  f32x16 o_acc[DB];
+#pragma unroll
+  for (int d = 0; d < DB; ++d) o_acc[d] = f32x16{};
+
+  const int kv_end = min(T, q0 + QB);
+
+  for (int kt = 0; kt < kv_end; kt += KB) {
+    // ---- stage K row-major (swizzled) + V transposed ----
+    for (int idx = threadIdx.x * 8; idx < KB * D; idx += 256 * 8) {
+      const int key = idx / D, d = idx % D;
+      const int kg = kt + key;
+      s16x8 kv8{}, vv8{};
+      if (kg < T) {
+        kv8 = *reinterpret_cast<const s16x8*>(&k[base + (long)kg * D + d]);
+        vv8 = *reinterpret_cast<const s16x8*>(&v[base + (long)kg * D + d]);
+      }
+      *reinterpret_cast<s16x8*>((char*)k_lds + swz(key, key * 256 + d * 2)) = kv8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) vt_lds[(d + e) * VT_STRIDE + key] = (uint16_t)vv8[e];
+    }
+    __syncthreads();
+
+    // ---- S = Q @ K^T (32x32 fp32 acc) ----
+    f32x16 s_acc{};
+#pragma unroll
+    for (int s = 0; s < KS; ++s) {
+      const int kk = s * 16 + 8 * hi;
+      bf16x8 k_frag = *reinterpret_cast<const bf16x8*>(
+          (char*)k_lds + swz(li, li * 256 + kk * 2));
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q_frag[s], k_frag, s_acc, 0, 0, 0);
+    }
+
+    // ---- online softmax per accumulator register (= per q row) ----
+    const int kj = kt + li;  // this lane's key column
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const int qi = qw + row;
+      float sv = s_acc[r] * scale + slope * (float)(kj - qi);
+      if (kj > qi || kj >= T || qi >= T) sv = NEG_INF;
+      const float mr = half_reduce_max(sv);
+      float p = 0.f, alpha = 1.f;
+      if (mr > NEG_INF * 0.5f) {
+        const float mn = fmaxf(m_run[r], mr);
+        alpha = expf(m_run[r] - mn);  // exp(-inf - mn) = 0 on first tile
+        m_run[r] = mn;
+        p = expf(sv - mn);
+      }
+      l_run[r] = l_run[r] * alpha + half_reduce_sum(p);
+#pragma unroll
+      for (int d = 0; d < DB; ++d) o_acc[d][r] *= alpha;
+      // dropout applies to the PV path only (scaled-mask, denominator keeps
+      // the full softmax — dropout acts on normalized probs, reference
+      // layers.py:174)
+      float p_pv = p;
+      if (p_drop > 0.f) {
+        const uint64_t idx = ((uint64_t)bh * (uint64_t)T + (uint64_t)qi) * (uint64_t)T + (uint64_t)kj;
+        p_pv = (uniform01(seed, idx) >= p_drop) ? p * inv_keep : 0.f;
+      }
+      p_lds[(wave * 32 + row) * VT_STRIDE + li] = f32_to_bf16(p_pv);
+    }
+
+    // ---- O += P @ V ----
+#pragma unroll
+    for (int d = 0; d < DB; ++d) {
+#pragma unroll
+      for (int s2 = 0; s2 < KB / 16; ++s2) {
+        const int kk = s2 * 16 + 8 * hi;
+        bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
+            &p_lds[(wave * 32 + li) * VT_STRIDE + kk]);
+        bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            &vt_lds[(d * 32 + li) * VT_STRIDE + kk]);
+        o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_frag, b_frag, o_acc[d], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: O / l, lse = m + log(l) ----
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const int qi = qw + row;
+    if (qi >= T) continue;
+    const float inv_l = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+#pragma unroll
+    for (int d = 0; d < DB; ++d)
+      o[base + (long)qi * D + d * 32 + li] = f32_to_bf16(o_acc[d][r] * inv_l);
+    if (li == 0)
+      lse[(long)bh * T + qi] = l_run[r] > 0.f ? m_run[r] + logf(l_run[r]) : NEG_INF;
+  }
+}
+
+template <int D>
+void launch_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+                const at::Tensor& slopes, at::Tensor& o, at::Tensor& lse, int B,
+                int H, int T, float scale, float p_drop, uint64_t seed,
+                hipStream_t stream) {
+  dim3 grid((T + QB - 1) / QB, B * H);
+  const size_t smem = (KB * 128 + D * VT_STRIDE + 4 * 32 * VT_STRIDE) * sizeof(uint16_t);
+  hipLaunchKernelGGL(flash_fwd_kernel<D>, grid, dim3(256), smem, stream,
+                     (const uint16_t*)q.data_ptr(), (const uint16_t*)k.data_ptr(),
+                     (const uint16_t*)v.data_ptr(), slopes.data_ptr<float>(),
+                     (uint16_t*)o.data_ptr(), lse.data_ptr<float>(), H, T, scale,
+                     p_drop, seed);
+}
+
+}  // namespace
+
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor slopes, double p_drop, int64_t seed) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.dim() == 4, "q must be (B,H,T,D) contiguous");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_fwd: bf16 only");
+  const int B = q.size(0), H = q.size(1), T = q.size(2), D = q.size(3);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, H, T}, q.options().dtype(at::kFloat));
+  auto sl = slopes.to(at::kFloat).contiguous();
+  const float scale = 1.0f / sqrtf((float)D);
+  auto stream = at::hip::getCurrentHIPStream();
+  switch (D) {
+    case 32: launch_fwd<32>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;
+    case 64: launch_fwd<64>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;
+    case 96: launch_fwd<96>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;
+    case 128: launch_fwd<128>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;
+    default: TORCH_CHECK(false, "attn_fwd: head_dim must be one of 32/64/96/128, got ", D);
+  }
+  return {o, lse};
+}

@@ -1,0 +1,34 @@
+// Python bindings for the zero_transformer_amd gfx950 HIP kernels.
+
+#include <torch/extension.h>
+
+#include <vector>
+
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, double eps);
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                      at::Tensor rstd, at::Tensor mean);
+at::Tensor gelu_fwd(at::Tensor x);
+at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x);
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets);
+at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                             at::Tensor dloss);
+void adamw_step(at::Tensor p, at::Tensor p_bf16, at::Tensor g, at::Tensor m,
+                at::Tensor v, long step, double lr, double beta1, double beta2,
+                double eps, double wd, double clip, double grad_scale);
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor slopes, double p_drop, int64_t seed);
+std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
+                                 at::Tensor v, at::Tensor slopes, at::Tensor o,
+                                 at::Tensor lse, double p_drop, int64_t seed);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd, "bias-free LayerNorm fwd (gfx950)");
+  m.def("layernorm_bwd", &layernorm_bwd, "bias-free LayerNorm bwd (gfx950)");
+  m.def("gelu_fwd", &gelu_fwd, "tanh GELU fwd (gfx950)");
+  m.def("gelu_bwd", &gelu_bwd, "tanh GELU bwd (gfx950)");
+  m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused gather CE fwd (gfx950)");
+  m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused gather CE bwd (gfx950)");
+  m.def("adamw_step", &adamw_step, "fused ZeRO-1 AdamW shard step (gfx950)");
+  m.def("attn_fwd", &attn_fwd, "fused causal ALiBi flash attention fwd (gfx950 MFMA)");
+  m.def("attn_bwd", &attn_bwd, "fused causal ALiBi flash attention bwd (gfx950 MFMA)");
+}

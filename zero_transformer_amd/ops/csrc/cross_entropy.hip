@@ -1,0 +1,117 @@
+// Fused gather-based cross entropy for gfx950: never materializes the
+// (N, 50304) one-hot of the reference (GPT.py:105-111, losses.py:10-23).
+// fp32 log-softmax statistics over bf16 logits.
+//
+// fwd: one block per row -> per-row loss (lse - logit[target]) + saved lse.
+// bwd: dlogits = (softmax - onehot) * dloss / N  in one vectorized pass.
+
+#include "common.h"
+
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+namespace {
+
+template <typename T>
+__global__ void ce_fwd_kernel(const T* __restrict__ logits,
+                              const long* __restrict__ targets,
+                              float* __restrict__ loss, float* __restrict__ lse_out,
+                              long rows, int V) {
+  __shared__ float scratch[16];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* lr = logits + row * V;
+    float m = -3.4e38f;
+    for (int c = threadIdx.x; c < V; c += blockDim.x) m = fmaxf(m, to_f32(lr[c]));
+    m = wave_reduce_max(m);
+    {  // cross-wave max via scratch
+      const int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+      if (lane == 0) scratch[wid] = m;
+      __syncthreads();
+      const int nw = blockDim.x / WAVE;
+      float v = (threadIdx.x < nw) ? scratch[threadIdx.x] : -3.4e38f;
+      if (wid == 0) v = wave_reduce_max(v);
+      if (threadIdx.x == 0) scratch[0] = v;
+      __syncthreads();
+      m = scratch[0];
+      __syncthreads();
+    }
+    float s = 0.f;
+    for (int c = threadIdx.x; c < V; c += blockDim.x) s += expf(to_f32(lr[c]) - m);
+    s = block_reduce_sum(s, scratch);
+    const float lse = m + logf(s);
+    if (threadIdx.x == 0) {
+      lse_out[row] = lse;
+      loss[row] = lse - to_f32(lr[targets[row]]);
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T>
+__global__ void ce_bwd_kernel(const T* __restrict__ logits,
+                              const long* __restrict__ targets,
+                              const float* __restrict__ lse,
+                              const float* __restrict__ dloss,  // scalar
+                              T* __restrict__ dlogits, long rows, int V) {
+  const float scale = dloss[0] / rows;  // mean over rows
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* lr = logits + row * V;
+    T* dr = dlogits + row * V;
+    const float l = lse[row];
+    const long t = targets[row];
+    for (int c = threadIdx.x; c < V; c += blockDim.x) {
+      float p = expf(to_f32(lr[c]) - l);
+      dr[c] = from_f32<T>(scale * (p - (c == t ? 1.f : 0.f)));
+    }
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets) {
+  TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() && logits.dim() == 2);
+  TORCH_CHECK(targets.scalar_type() == at::kLong);
+  const long rows = logits.size(0);
+  const int V = logits.size(1);
+  auto loss = at::empty({rows}, logits.options().dtype(at::kFloat));
+  auto lse = at::empty({rows}, logits.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  const int block = 256;
+  const int grid = int(std::min<long>(rows, 2048));
+  if (logits.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(ce_fwd_kernel<uint16_t>, dim3(grid), dim3(block), 0, stream,
+                       (const uint16_t*)logits.data_ptr(), targets.data_ptr<long>(),
+                       loss.data_ptr<float>(), lse.data_ptr<float>(), rows, V);
+  } else if (logits.scalar_type() == at::kFloat) {
+    hipLaunchKernelGGL(ce_fwd_kernel<float>, dim3(grid), dim3(block), 0, stream,
+                       logits.data_ptr<float>(), targets.data_ptr<long>(),
+                       loss.data_ptr<float>(), lse.data_ptr<float>(), rows, V);
+  } else {
+    TORCH_CHECK(false, "cross_entropy: unsupported dtype");
+  }
+  return {loss.mean(), lse};
+}
+
+at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                             at::Tensor dloss) {
+  const long rows = logits.size(0);
+  const int V = logits.size(1);
+  auto dlogits = at::empty_like(logits);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto dl = dloss.to(at::kFloat).contiguous();
+  const int block = 256;
+  const int grid = int(std::min<long>(rows, 2048));
+  if (logits.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(ce_bwd_kernel<uint16_t>, dim3(grid), dim3(block), 0, stream,
+                       (const uint16_t*)logits.data_ptr(), targets.data_ptr<long>(),
+                       lse.data_ptr<float>(), dl.data_ptr<float>(),
+                       (uint16_t*)dlogits.data_ptr(), rows, V);
+  } else {
+    hipLaunchKernelGGL(ce_bwd_kernel<float>, dim3(grid), dim3(block), 0, stream,
+                       logits.data_ptr<float>(), targets.data_ptr<long>(),
+                       lse.data_ptr<float>(), dl.data_ptr<float>(),
+                       dlogits.data_ptr<float>(), rows, V);
+  }
+  return dlogits;
+}

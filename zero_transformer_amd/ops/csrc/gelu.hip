@@ -1,0 +1,117 @@
+// tanh-approximate GELU forward/backward (flax nn.gelu default,
+// reference layers.py:68). Memory-bound elementwise: vectorized 8x bf16
+// access per lane (G13), grid-stride.
+
+#include "common.h"
+
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+namespace {
+
+constexpr float kC0 = 0.7978845608028654f;  // sqrt(2/pi)
+constexpr float kC1 = 0.044715f;
+
+ZTA_DEV float gelu_f(float x) {
+  float t = tanhf(kC0 * (x + kC1 * x * x * x));
+  return 0.5f * x * (1.f + t);
+}
+
+ZTA_DEV float gelu_grad_f(float x) {
+  float x2 = x * x;
+  float u = kC0 * (x + kC1 * x * x2);
+  float t = tanhf(u);
+  float sech2 = 1.f - t * t;
+  return 0.5f * (1.f + t) + 0.5f * x * sech2 * kC0 * (1.f + 3.f * kC1 * x2);
+}
+
+// bf16: 8 elements per lane per iteration (16B vector load/store).
+__global__ void gelu_fwd_bf16(const s16x8* __restrict__ x, s16x8* __restrict__ y,
+                              long n8) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    s16x8 v = x[i];
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = (short)f32_to_bf16(gelu_f(bf16_to_f32((uint16_t)v[j])));
+    y[i] = o;
+  }
+}
+
+__global__ void gelu_bwd_bf16(const s16x8* __restrict__ dy, const s16x8* __restrict__ x,
+                              s16x8* __restrict__ dx, long n8) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    s16x8 g = dy[i], v = x[i];
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = (short)f32_to_bf16(bf16_to_f32((uint16_t)g[j]) *
+                                gelu_grad_f(bf16_to_f32((uint16_t)v[j])));
+    dx[i] = o;
+  }
+}
+
+template <typename T>
+__global__ void gelu_fwd_scalar(const T* __restrict__ x, T* __restrict__ y, long n) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    y[i] = from_f32<T>(gelu_f(to_f32(x[i])));
+}
+
+template <typename T>
+__global__ void gelu_bwd_scalar(const T* __restrict__ dy, const T* __restrict__ x,
+                                T* __restrict__ dx, long n) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    dx[i] = from_f32<T>(to_f32(dy[i]) * gelu_grad_f(to_f32(x[i])));
+}
+
+}  // namespace
+
+at::Tensor gelu_fwd(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  auto y = at::empty_like(x);
+  auto stream = at::hip::getCurrentHIPStream();
+  const long n = x.numel();
+  const int block = 256;
+  if (x.scalar_type() == at::kBFloat16 && n % 8 == 0) {
+    const long n8 = n / 8;
+    hipLaunchKernelGGL(gelu_fwd_bf16, dim3(capped_grid(n8, block)), dim3(block), 0,
+                       stream, (const s16x8*)x.data_ptr(), (s16x8*)y.data_ptr(), n8);
+  } else if (x.scalar_type() == at::kFloat) {
+    hipLaunchKernelGGL(gelu_fwd_scalar<float>, dim3(capped_grid(n, block)), dim3(block),
+                       0, stream, x.data_ptr<float>(), y.data_ptr<float>(), n);
+  } else if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(gelu_fwd_scalar<uint16_t>, dim3(capped_grid(n, block)),
+                       dim3(block), 0, stream, (const uint16_t*)x.data_ptr(),
+                       (uint16_t*)y.data_ptr(), n);
+  } else {
+    TORCH_CHECK(false, "gelu: unsupported dtype");
+  }
+  return y;
+}
+
+at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x) {
+  auto dx = at::empty_like(x);
+  auto stream = at::hip::getCurrentHIPStream();
+  const long n = x.numel();
+  const int block = 256;
+  if (x.scalar_type() == at::kBFloat16 && n % 8 == 0) {
+    const long n8 = n / 8;
+    hipLaunchKernelGGL(gelu_bwd_bf16, dim3(capped_grid(n8, block)), dim3(block), 0,
+                       stream, (const s16x8*)dy.data_ptr(), (const s16x8*)x.data_ptr(),
+                       (s16x8*)dx.data_ptr(), n8);
+  } else if (x.scalar_type() == at::kFloat) {
+    hipLaunchKernelGGL(gelu_bwd_scalar<float>, dim3(capped_grid(n, block)), dim3(block),
+                       0, stream, dy.data_ptr<float>(), x.data_ptr<float>(),
+                       dx.data_ptr<float>(), n);
+  } else {
+    hipLaunchKernelGGL(gelu_bwd_scalar<uint16_t>, dim3(capped_grid(n, block)),
+                       dim3(block), 0, stream, (const uint16_t*)dy.data_ptr(),
+                       (const uint16_t*)x.data_ptr(), (uint16_t*)dx.data_ptr(), n);
+  }
+  return dx;
+}
