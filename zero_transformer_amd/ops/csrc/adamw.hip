@@ -9,6 +9,7 @@
 
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <torch/extension.h>
 
 namespace {
@@ -45,7 +46,7 @@ void adamw_step(at::Tensor p, at::Tensor p_bf16, at::Tensor g, at::Tensor m,
   TORCH_CHECK(m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
   const long n = p.numel();
   const bool emit = p_bf16.numel() == n;
-  auto stream = at::hip::getCurrentHIPStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int block = 256;
   const int grid = capped_grid(n, block, 4096);
   const float inv_bc1 = 1.f / (1.f - powf((float)beta1, (float)step));

@@ -20,6 +20,7 @@
 
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <torch/extension.h>
 
 namespace {
@@ -353,7 +354,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   auto delta = at::empty({B, H, T}, q.options().dtype(at::kFloat));
   auto sl = slopes.to(at::kFloat).contiguous();
   const float scale = 1.0f / sqrtf((float)D);
-  auto stream = at::hip::getCurrentHIPStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   {
     const long rows = (long)B * H * T;
     const int block = 256;

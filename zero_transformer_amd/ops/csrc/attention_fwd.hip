@@ -24,6 +24,7 @@
 
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <torch/extension.h>
 
 namespace {
@@ -207,7 +208,7 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto lse = at::empty({B, H, T}, q.options().dtype(at::kFloat));
   auto sl = slopes.to(at::kFloat).contiguous();
   const float scale = 1.0f / sqrtf((float)D);
-  auto stream = at::hip::getCurrentHIPStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   switch (D) {
     case 32: launch_fwd<32>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;
     case 64: launch_fwd<64>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;

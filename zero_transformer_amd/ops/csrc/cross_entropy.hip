@@ -9,6 +9,7 @@
 
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <torch/extension.h>
 
 namespace {
@@ -76,7 +77,7 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets)
   const int V = logits.size(1);
   auto loss = at::empty({rows}, logits.options().dtype(at::kFloat));
   auto lse = at::empty({rows}, logits.options().dtype(at::kFloat));
-  auto stream = at::hip::getCurrentHIPStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int block = 256;
   const int grid = int(std::min<long>(rows, 2048));
   if (logits.scalar_type() == at::kBFloat16) {
@@ -98,7 +99,7 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor l
   const long rows = logits.size(0);
   const int V = logits.size(1);
   auto dlogits = at::empty_like(logits);
-  auto stream = at::hip::getCurrentHIPStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   auto dl = dloss.to(at::kFloat).contiguous();
   const int block = 256;
   const int grid = int(std::min<long>(rows, 2048));

@@ -6,6 +6,7 @@
 
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <torch/extension.h>
 
 namespace {
@@ -74,7 +75,7 @@ __global__ void gelu_bwd_scalar(const T* __restrict__ dy, const T* __restrict__ 
 at::Tensor gelu_fwd(at::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
   auto y = at::empty_like(x);
-  auto stream = at::hip::getCurrentHIPStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const long n = x.numel();
   const int block = 256;
   if (x.scalar_type() == at::kBFloat16 && n % 8 == 0) {
@@ -96,7 +97,7 @@ at::Tensor gelu_fwd(at::Tensor x) {
 
 at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x) {
   auto dx = at::empty_like(x);
-  auto stream = at::hip::getCurrentHIPStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const long n = x.numel();
   const int block = 256;
   if (x.scalar_type() == at::kBFloat16 && n % 8 == 0) {

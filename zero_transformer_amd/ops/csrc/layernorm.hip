@@ -7,6 +7,7 @@
 
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <torch/extension.h>
 
 namespace {
@@ -93,7 +94,7 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, double eps) {
   auto y = at::empty_like(x);
   auto mean = at::empty({rows}, x.options().dtype(at::kFloat));
   auto rstd = at::empty({rows}, x.options().dtype(at::kFloat));
-  auto stream = at::hip::getCurrentHIPStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int block = 256;
   const int grid = int(std::min<long>(rows, 2048));
   if (x.scalar_type() == at::kBFloat16) {
@@ -117,7 +118,7 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   const long rows = x.numel() / C;
   auto dx = at::empty_like(x);
   auto dw_f32 = at::zeros({C}, x.options().dtype(at::kFloat));
-  auto stream = at::hip::getCurrentHIPStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int block = 256;
   const int grid = int(std::min<long>(rows, 1024));
   const size_t smem = (C + 16) * sizeof(float);
