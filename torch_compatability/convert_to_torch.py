@@ -1,0 +1,48 @@
+"""Convert a training checkpoint (params_<step>.pt, full fp32 state dict in
+the .pth key layout) into an inference-model .pth, with optional vocab
+truncation — the role of the reference's flax_to_pytorch.py + convert_to_torch.py
+(match_and_save truncates the padded 50304 vocab to the model's, :96-114).
+
+    python torch_compatability/convert_to_torch.py \
+        --checkpoint checkpoints/760m/params_82000.pt \
+        --model-size 760m --out checkpoints/torch_760m.pth
+"""
+
+from __future__ import annotations
+
+import argparse
+
+import torch
+
+
+def match_and_save(model, checkpoint_path: str, out_save_path: str) -> None:
+    """Load trainer params into `model` (truncating vocab rows) and save its
+    state_dict (.pth contract)."""
+    sd = torch.load(checkpoint_path, map_location="cpu", weights_only=True)
+    out = {}
+    for k, v in sd.items():
+        if k in ("wte.weight", "lm_head.weight"):
+            v = v[: model.vocab_size]
+        out[k] = v.float()
+    if "lm_head.weight" not in out and "wte.weight" in out:
+        out["lm_head.weight"] = out["wte.weight"].clone()
+    model.load_state_dict(out)
+    torch.save(model.state_dict(), out_save_path)
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--checkpoint", required=True)
+    p.add_argument("--model-size", required=True)
+    p.add_argument("--model-cfg", default="torch_compatability/model_config.yaml")
+    p.add_argument("--out", required=True)
+    args = p.parse_args()
+    from torch_compatability.GPT2 import model_getter
+
+    model = model_getter(args.model_size, config_path=args.model_cfg)
+    match_and_save(model, args.checkpoint, args.out)
+    print(f"saved {args.out}")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,30 @@
+"""Extract raw model params from a full training checkpoint directory into a
+standalone .pth (the role of the reference's extract_msgpack.py:28-47, which
+pulled params out of a Flax train-state checkpoint).
+
+    python torch_compatability/extract_params.py --workdir checkpoints/760m \
+        [--step 82000] --out params_raw.pth
+"""
+
+from __future__ import annotations
+
+import argparse
+
+import torch
+
+from zero_transformer_amd.utils import checkpoint as ckpt
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--workdir", required=True)
+    p.add_argument("--step", type=int, default=None)
+    p.add_argument("--out", required=True)
+    args = p.parse_args()
+    params, _, step = ckpt.restore_checkpoint(args.workdir, args.step)
+    torch.save(params, args.out)
+    print(f"extracted step {step} params -> {args.out}")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,200 @@
+"""KV-cached inference GPT (the torch_compatability sidecar model).
+
+Same capability as the reference's PyTorch inference mirror
+(torch_compatability/GPT2.py:49-474): ALiBi attention via
+scaled_dot_product_attention with an additive mask, per-layer KV cache
+(concat along the sequence axis), dynamic ALiBi mask rebuild when the
+context grows, weight-tied lm_head, greedy/sampling generate, shifted CE
+loss. Loads / saves the .pth state-dict contract
+(flax_to_pytorch.py:10-35,96-114 key layout).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops.reference import alibi_slopes
+from ..utils.config import load_config
+
+
+def _alibi_bias(slopes: torch.Tensor, Tq: int, Tk: int, device, dtype) -> torch.Tensor:
+    """Additive (H, Tq, Tk) mask: ALiBi bias + causal -inf.
+
+    For decode (Tq < Tk) the q rows are the LAST Tq positions of the Tk
+    context (reference GPT2.py:193-235: decode path uses the last-row mask).
+    """
+    H = slopes.shape[0]
+    qpos = torch.arange(Tk - Tq, Tk, device=device, dtype=torch.float32)
+    kpos = torch.arange(Tk, device=device, dtype=torch.float32)
+    rel = kpos.view(1, Tk) - qpos.view(Tq, 1)  # j - i
+    bias = slopes.to(device).float().view(H, 1, 1) * rel.view(1, Tq, Tk)
+    bias = bias.masked_fill(rel.view(1, Tq, Tk) > 0, float("-inf"))
+    return bias.to(dtype)
+
+
+class InferenceAttention(nn.Module):
+    def __init__(self, dim: int, heads: int, alibi: bool = True):
+        super().__init__()
+        self.num_head = heads
+        self.head_dim = dim // heads
+        self.query = nn.Linear(dim, dim, bias=False)
+        self.key = nn.Linear(dim, dim, bias=False)
+        self.value = nn.Linear(dim, dim, bias=False)
+        self.fc_resid = nn.Linear(dim, dim, bias=False)
+        if alibi:
+            self.register_buffer("slopes", alibi_slopes(heads), persistent=False)
+        else:
+            self.slopes = None
+
+    def forward(
+        self,
+        x: torch.Tensor,
+        layer_past: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
+        use_cache: bool = False,
+    ):
+        B, T, C = x.shape
+        H, D = self.num_head, self.head_dim
+        q = self.query(x).view(B, T, H, D).transpose(1, 2)
+        k = self.key(x).view(B, T, H, D).transpose(1, 2)
+        v = self.value(x).view(B, T, H, D).transpose(1, 2)
+        if layer_past is not None:
+            pk, pv = layer_past
+            k = torch.cat([pk, k], dim=2)  # concat along seq (GPT2.py:177-182)
+            v = torch.cat([pv, v], dim=2)
+        present = (k, v) if use_cache else None
+        Tk = k.shape[2]
+        if self.slopes is not None:
+            mask = _alibi_bias(self.slopes, T, Tk, x.device, q.dtype).unsqueeze(0)
+            out = F.scaled_dot_product_attention(q, k, v, attn_mask=mask)
+        else:
+            if T == Tk:
+                out = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+            else:
+                causal = torch.zeros(T, Tk, device=x.device, dtype=q.dtype)
+                rel = torch.arange(Tk, device=x.device).view(1, Tk) - torch.arange(
+                    Tk - T, Tk, device=x.device
+                ).view(T, 1)
+                causal = causal.masked_fill(rel > 0, float("-inf"))
+                out = F.scaled_dot_product_attention(q, k, v, attn_mask=causal.view(1, 1, T, Tk))
+        out = out.transpose(1, 2).reshape(B, T, C)
+        return self.fc_resid(out), present
+
+
+class InferenceMLP(nn.Module):
+    def __init__(self, dim: int):
+        super().__init__()
+        self.fc1 = nn.Linear(dim, 4 * dim, bias=False)
+        self.fc_resid = nn.Linear(4 * dim, dim, bias=False)
+
+    def forward(self, x):
+        return self.fc_resid(F.gelu(self.fc1(x), approximate="tanh"))
+
+
+class InferenceBlock(nn.Module):
+    def __init__(self, dim: int, heads: int, alibi: bool = True):
+        super().__init__()
+        self.ln1 = nn.LayerNorm(dim, elementwise_affine=True, bias=False, eps=1e-6)
+        self.attn = InferenceAttention(dim, heads, alibi)
+        self.ln2 = nn.LayerNorm(dim, elementwise_affine=True, bias=False, eps=1e-6)
+        self.mlp = InferenceMLP(dim)
+
+    def forward(self, x, layer_past=None, use_cache=False):
+        a, present = self.attn(self.ln1(x), layer_past, use_cache)
+        x = x + a
+        x = x + self.mlp(self.ln2(x))
+        return x, present
+
+
+class GPT2(nn.Module):
+    """Inference model (reference torch_compatability/GPT2.py:297-445)."""
+
+    def __init__(self, embedding_dim: int, vocab_size: int, num_head: int,
+                 num_ctx: int, N: int, alibi: bool = True):
+        super().__init__()
+        self.N = N
+        self.vocab_size = vocab_size
+        self.num_ctx = num_ctx
+        self.wte = nn.Embedding(vocab_size, embedding_dim)
+        self.blocks = nn.ModuleList(
+            [InferenceBlock(embedding_dim, num_head, alibi) for _ in range(N)]
+        )
+        self.norm = nn.LayerNorm(embedding_dim, elementwise_affine=True, bias=False, eps=1e-6)
+        self.lm_head = nn.Linear(embedding_dim, vocab_size, bias=False)
+        self.lm_head.weight = self.wte.weight  # tied (GPT2.py:350)
+
+    def forward(
+        self,
+        idx: torch.Tensor,
+        labels: Optional[torch.Tensor] = None,
+        use_cache: bool = False,
+        past_states: Optional[List] = None,
+    ):
+        x = self.wte(idx)
+        presents = [] if use_cache else None
+        if past_states is None:
+            past_states = [None] * self.N
+        for block, past in zip(self.blocks, past_states):
+            x, present = block(x, past, use_cache)
+            if use_cache:
+                presents.append(present)
+        x = self.norm(x)
+        logits = self.lm_head(x)
+        if labels is not None:
+            tgt = labels[..., 1:].reshape(-1)
+            lg = logits[..., :-1, :].reshape(-1, logits.shape[-1])
+            loss = F.cross_entropy(lg.float(), tgt)
+            return logits, loss
+        if use_cache:
+            return logits, presents
+        return logits
+
+    @torch.no_grad()
+    def generate(
+        self,
+        idx: torch.Tensor,
+        max_new_tokens: int,
+        temperature: float = 1.0,
+        sample: bool = False,
+        top_k: Optional[int] = None,
+    ) -> torch.Tensor:
+        """Simple no-cache generate (reference GPT2.py:354-400, lm-eval path)."""
+        for _ in range(max_new_tokens):
+            ctx = idx[:, -self.num_ctx:]
+            logits = self.forward(ctx)
+            logits = logits[:, -1, :] / max(temperature, 1e-5)
+            if top_k is not None:
+                v, _ = torch.topk(logits, min(top_k, logits.size(-1)))
+                logits[logits < v[:, [-1]]] = float("-inf")
+            probs = F.softmax(logits.float(), dim=-1)
+            nxt = torch.multinomial(probs, 1) if sample else probs.argmax(-1, keepdim=True)
+            idx = torch.cat([idx, nxt], dim=1)
+        return idx
+
+
+def model_getter(
+    model_size: str,
+    config_path: str = "torch_compatability/model_config.yaml",
+    model_checkpoint: Optional[str] = None,
+) -> GPT2:
+    """Build (and optionally load) an inference model from the YAML zoo
+    (reference GPT2.py:448-474)."""
+    configs = load_config(config_path)
+    assert model_size in configs, "Invalid model size provided"
+    c = configs[model_size]
+    model = GPT2(
+        embedding_dim=c.embedding_dim,
+        vocab_size=c.vocab_size,
+        num_head=c.num_head,
+        num_ctx=c.num_ctx,
+        N=c.N,
+        alibi=bool(c.get("alibi_attn", True)),
+    )
+    if model_checkpoint is not None:
+        sd = torch.load(model_checkpoint, map_location="cpu", weights_only=True)
+        model.load_state_dict(sd)
+    return model
