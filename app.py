@@ -1,0 +1,121 @@
+"""Interactive inference app (the reference app.py role, MI355X-native).
+
+Two modes:
+  CLI:    python app.py --model-size 760m --checkpoint torch_760m.pth \
+              --prompt "Hello" --max-new-tokens 64 --top-p 0.95
+  Server: python app.py --model-size 760m --checkpoint ... --serve --port 7860
+          POST /generate {"prompt": ..., "max_new_tokens": ..., ...}
+          (FastAPI + uvicorn instead of the reference's Gradio UI — same
+          streaming KV-cached generation + samplers, reference app.py:42-142.)
+
+Tokenizer: GPT-NeoX-20B via transformers when its assets are cached locally
+(reference app.py:27); otherwise a byte-level fallback so the app works on
+network-less boxes and with the 256-vocab test model.
+"""
+
+from __future__ import annotations
+
+import argparse
+
+import torch
+
+from torch_compatability.GPT2 import model_getter
+from zero_transformer_amd.models.sampling import generate_stream
+
+
+class ByteTokenizer:
+    eos_token_id = None
+
+    def encode(self, s: str):
+        return list(s.encode("utf-8", errors="replace"))
+
+    def decode(self, ids):
+        return bytes(int(i) % 256 for i in ids).decode("utf-8", errors="replace")
+
+
+def load_tokenizer():
+    try:
+        from transformers import AutoTokenizer
+
+        return AutoTokenizer.from_pretrained("EleutherAI/gpt-neox-20b", local_files_only=True)
+    except Exception:
+        return ByteTokenizer()
+
+
+def model_creator(size: str, checkpoint: str | None, device: torch.device):
+    """reference app.py:30-39: build -> device -> half -> eval."""
+    model = model_getter(size, model_checkpoint=checkpoint)
+    model = model.to(device)
+    if device.type == "cuda":
+        model = model.half()
+    return model.eval()
+
+
+def generate_text(model, tok, device, prompt: str, **kw) -> str:
+    ids = tok.encode(prompt)
+    idx = torch.tensor([ids], dtype=torch.long, device=device)
+    out = list(generate_stream(model, idx, eos_token=getattr(tok, "eos_token_id", None), **kw))
+    return tok.decode(out)
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--model-size", default="test")
+    p.add_argument("--checkpoint", default=None)
+    p.add_argument("--prompt", default=None)
+    p.add_argument("--max-new-tokens", type=int, default=128)
+    p.add_argument("--temperature", type=float, default=0.8)
+    p.add_argument("--top-k", type=int, default=0)
+    p.add_argument("--top-p", type=float, default=0.95)
+    p.add_argument("--repetition-penalty", type=float, default=1.1)
+    p.add_argument("--greedy", action="store_true")
+    p.add_argument("--serve", action="store_true")
+    p.add_argument("--port", type=int, default=7860)
+    args = p.parse_args()
+
+    device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
+    model = model_creator(args.model_size, args.checkpoint, device)
+    tok = load_tokenizer()
+    kw = dict(
+        max_new_tokens=args.max_new_tokens,
+        temperature=args.temperature,
+        top_k=args.top_k,
+        top_p=args.top_p,
+        repetition_penalty=args.repetition_penalty,
+        sample=not args.greedy,
+    )
+
+    if args.serve:
+        from fastapi import FastAPI
+        from pydantic import BaseModel
+        import uvicorn
+
+        app = FastAPI(title="zero_transformer_amd inference")
+
+        class Req(BaseModel):
+            prompt: str
+            max_new_tokens: int = 128
+            temperature: float = 0.8
+            top_k: int = 0
+            top_p: float = 0.95
+            repetition_penalty: float = 1.1
+            greedy: bool = False
+
+        @app.post("/generate")
+        def generate(req: Req):
+            text = generate_text(
+                model, tok, device, req.prompt,
+                max_new_tokens=req.max_new_tokens, temperature=req.temperature,
+                top_k=req.top_k, top_p=req.top_p,
+                repetition_penalty=req.repetition_penalty, sample=not req.greedy,
+            )
+            return {"completion": text}
+
+        uvicorn.run(app, host="0.0.0.0", port=args.port)
+    else:
+        prompt = args.prompt or "Hello"
+        print(prompt + generate_text(model, tok, device, prompt, **kw))
+
+
+if __name__ == "__main__":
+    main()

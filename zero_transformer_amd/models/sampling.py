@@ -1,0 +1,78 @@
+"""Sampling utilities + streaming KV-cached generation
+(reference app.py:42-142: repetition penalty, temperature, top-k, nucleus,
+greedy, EOS stop, per-token streaming with layer_past cache)."""
+
+from __future__ import annotations
+
+from typing import Iterator, List, Optional
+
+import torch
+import torch.nn.functional as F
+
+
+def apply_repetition_penalty(
+    logits: torch.Tensor, generated: List[int], penalty: float
+) -> torch.Tensor:
+    """CTRL-style repetition penalty (reference app.py:97-108)."""
+    if penalty == 1.0 or not generated:
+        return logits
+    idx = torch.tensor(sorted(set(generated)), device=logits.device)
+    sel = logits[..., idx]
+    sel = torch.where(sel > 0, sel / penalty, sel * penalty)
+    logits[..., idx] = sel
+    return logits
+
+
+def top_k_filter(logits: torch.Tensor, k: int) -> torch.Tensor:
+    """Keep the k highest logits (reference app.py:111-115)."""
+    if k <= 0:
+        return logits
+    v, _ = torch.topk(logits, min(k, logits.size(-1)))
+    return logits.masked_fill(logits < v[..., -1, None], float("-inf"))
+
+
+def top_p_filter(logits: torch.Tensor, p: float) -> torch.Tensor:
+    """Nucleus filtering (reference app.py:118-142)."""
+    if p >= 1.0:
+        return logits
+    sorted_logits, sorted_idx = torch.sort(logits, descending=True)
+    cum = torch.cumsum(F.softmax(sorted_logits.float(), dim=-1), dim=-1)
+    remove = cum > p
+    remove[..., 1:] = remove[..., :-1].clone()
+    remove[..., 0] = False
+    mask = remove.scatter(-1, sorted_idx, remove)
+    return logits.masked_fill(mask, float("-inf"))
+
+
+@torch.no_grad()
+def generate_stream(
+    model,
+    idx: torch.Tensor,
+    max_new_tokens: int = 128,
+    temperature: float = 1.0,
+    top_k: int = 0,
+    top_p: float = 1.0,
+    repetition_penalty: float = 1.0,
+    sample: bool = True,
+    eos_token: Optional[int] = None,
+) -> Iterator[int]:
+    """Streaming generation with KV cache (reference app.py:42-94)."""
+    model.eval()
+    generated: List[int] = idx[0].tolist()
+    logits, states = model(idx, use_cache=True)
+    for _ in range(max_new_tokens):
+        lg = logits[:, -1, :].clone()
+        lg = apply_repetition_penalty(lg, generated, repetition_penalty)
+        lg = lg / max(temperature, 1e-5)
+        if top_k:
+            lg = top_k_filter(lg, top_k)
+        if top_p < 1.0:
+            lg = top_p_filter(lg, top_p)
+        probs = F.softmax(lg.float(), dim=-1)
+        nxt = int(torch.multinomial(probs, 1).item()) if sample else int(probs.argmax().item())
+        if eos_token is not None and nxt == eos_token:
+            return
+        generated.append(nxt)
+        yield nxt
+        cur = torch.tensor([[nxt]], device=idx.device)
+        logits, states = model(cur, use_cache=True, past_states=states)
