@@ -76,7 +76,8 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
   const int li = lane & 31, hi = lane >> 5;
   const int qw = q0 + wave * 32;
   const float slope = slopes[h];
-  const float inv_keep = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
+  const uint32_t drop_thr = (uint32_t)(p_drop * 65536.0f + 0.5f);
+  const float inv_keep = drop_thr ? 65536.0f / (65536.0f - (float)drop_thr) : 1.0f;
 
   constexpr int KS = D / 16;
   constexpr int DB = D / 32;
@@ -142,11 +143,12 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
       const int qi = qw + row;
       float p = 0.f;
       if (!(kj > qi || kj >= T || qi >= T))
-        p = expf(s_acc[r] * scale + slope * (float)(kj - qi) - lse_r[r]);
+        p = __expf(s_acc[r] * scale + slope * (float)(kj - qi) - lse_r[r]);
       float dp = dp_acc[r];
-      if (p_drop > 0.f) {
-        const uint64_t idx = ((uint64_t)bh * (uint64_t)T + (uint64_t)qi) * (uint64_t)T + (uint64_t)kj;
-        dp = (uniform01(seed, idx) >= p_drop) ? dp * inv_keep : 0.f;
+      if (drop_thr) {
+        const uint64_t bits = drop_bits(seed, bh * T + qi, kj >> 2);
+        const bool keep = (uint16_t)(bits >> (16 * (kj & 3))) >= drop_thr;
+        dp = keep ? dp * inv_keep : 0.f;
       }
       const float ds = scale * p * (dp - delta_r[r]);
       ds_lds[(wave * 32 + row) * TSTRIDE + li] = f32_to_bf16(ds);
@@ -205,7 +207,8 @@ __global__ __launch_bounds__(256) void flash_dkdv_kernel(
   const int li = lane & 31, hi = lane >> 5;
   const int kw = k0 + wave * 32;  // this wave's first key
   const float slope = slopes[h];
-  const float inv_keep = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
+  const uint32_t drop_thr = (uint32_t)(p_drop * 65536.0f + 0.5f);
+  const float inv_keep = drop_thr ? 65536.0f / (65536.0f - (float)drop_thr) : 1.0f;
 
   constexpr int KS = D / 16;
   constexpr int DB = D / 32;
@@ -272,12 +275,12 @@ __global__ __launch_bounds__(256) void flash_dkdv_kernel(
       const int kj = kw + row;
       float p = 0.f;
       if (!(kj > qi || kj >= T || qi >= T))
-        p = expf(st_acc[r] * scale + slope * (float)(kj - qi) - lse_q);
+        p = __expf(st_acc[r] * scale + slope * (float)(kj - qi) - lse_q);
       float dp = dpt_acc[r];
       float p_pv = p;
-      if (p_drop > 0.f) {
-        const uint64_t idx = ((uint64_t)bh * (uint64_t)T + (uint64_t)qi) * (uint64_t)T + (uint64_t)kj;
-        const bool keep = uniform01(seed, idx) >= p_drop;
+      if (drop_thr) {
+        const uint64_t bits = drop_bits(seed, bh * T + qi, kj >> 2);
+        const bool keep = (uint16_t)(bits >> (16 * (kj & 3))) >= drop_thr;
         dp = keep ? dp * inv_keep : 0.f;
         p_pv = keep ? p * inv_keep : 0.f;
       }

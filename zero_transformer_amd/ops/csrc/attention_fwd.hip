@@ -2,23 +2,27 @@
 //
 // Replaces the reference's dense O(T^2) materialized attention
 // (layers.py:159-178: q@k^T, +alibi, tril mask, fp32 softmax, dropout, @v)
-// with a flash-style online-softmax kernel: per 128-row Q block, iterate
-// 32-key K/V tiles with running (m, l) in fp32 — the softmax statistics stay
-// fp32 end to end (the bf16-softmax failure of logs/580.md:94-98 cannot
-// happen by construction). ALiBi is applied as the true bias
-// slope*(j - i) (shift-equivalent to the reference's single-row mask,
-// layers.py:33-44). Dropout uses a counter-based RNG so backward
-// regenerates the identical mask without storing it.
+// with a flash-style online-softmax kernel; softmax statistics stay fp32 end
+// to end (the bf16-softmax failure of logs/580.md:94-98 cannot happen by
+// construction). ALiBi is the true bias slope*(j-i) (shift-equivalent to the
+// reference's single-row mask, layers.py:33-44). Dropout uses a counter RNG
+// so backward regenerates the identical mask without storing it.
 //
-// Structure: 4 waves/block, each wave owns 32 q rows via one
-// v_mfma_f32_32x32x16_bf16 accumulator; K staged row-major in XOR-swizzled
-// LDS (T2: conflict-free ds_read_b128 B-fragments), V staged transposed
-// ([d][key]) so the PV B-operand reads are k-contiguous.
+// Structure (the CDNA4 guide's swapped-QK^T recipe, §B attention):
+//   * 8 waves/block, each owning 32 q rows (QB = 256); K/V staged once per
+//     64-key tile, two 32-key MFMA subtiles per staging.
+//   * QK^T is computed SWAPPED — mfma(A=K, B=Q^T) — so the C column index
+//     is the q row: each lane holds a full P-row segment in registers and
+//     the softmax row-reduce is 15 in-lane ops + one shfl_xor(32), with
+//     scalar (not per-reg) running m/l. No P round-trip through LDS:
+//     P -> bf16 A-fragments via v_cvt_pk_bf16_f32 + v_permlane32_swap (T12).
+//   * K staged row-major in XOR-swizzled LDS (T2, conflict-free
+//     ds_read_b128); V transposed ([d][key]) so PV B-reads are k-contiguous.
+//   * defer-max: the O rescale is skipped when no lane's running max grew
+//     (exact: threshold 0), __expf (v_exp_f32) throughout.
 //
-// Fragment maps (gfx950, §3 of the CDNA4 guide):
-//   A: lane l holds A[i = l%32][k = 8*(l/32) + e], e = 0..7
-//   B: lane l holds B[k = 8*(l/32) + e][j = l%32]
-//   C/D: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+// Fragment maps (gfx950, §3): A: lane l holds A[i=l%32][k=8*(l/32)+e];
+// B: B[k=8*(l/32)+e][j=l%32]; C/D: col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5).
 
 #include "common.h"
 
@@ -31,29 +35,23 @@ namespace {
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
-constexpr int QB = 128;  // q rows per block (4 waves x 32)
-constexpr int KB = 32;   // kv tile
+constexpr int NW = 8;        // waves per block
+constexpr int QB = NW * 32;  // q rows per block
+constexpr int KB = 64;       // staged kv tile (2 x 32-key subtiles)
 constexpr float NEG_INF = -3.0e38f;
+constexpr int VT_STRIDE = KB + 8;  // transposed-V row stride (elements)
 
-// XOR swizzle for 256-byte-stride LDS rows (T2): spreads ds_read_b128 lane
-// groups over 8 slots. Applied identically on write and read.
 ZTA_DEV int swz(int row, int byte_off) { return byte_off ^ ((row & 7) << 4); }
 
-// K tile: [KB][128] bf16, row stride 256 B, swizzled.
-// Vt tile: [D][KB + 8] bf16, row stride 80 B (bank-conflict-free without swizzle).
-// P tile (per wave): [32][KB + 8] bf16.
-constexpr int VT_STRIDE = KB + 8;  // elements
-
 template <int D>
-__global__ __launch_bounds__(256) void flash_fwd_kernel(
+__global__ __launch_bounds__(512) void flash_fwd_kernel(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const float* __restrict__ slopes,
     uint16_t* __restrict__ o, float* __restrict__ lse, int H, int T,
     float scale, float p_drop, uint64_t seed) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  uint16_t* k_lds = (uint16_t*)smem;                     // KB*128
-  uint16_t* vt_lds = k_lds + KB * 128;                   // D*VT_STRIDE
-  uint16_t* p_lds = vt_lds + D * VT_STRIDE;              // 4*32*VT_STRIDE
+  uint16_t* k_lds = (uint16_t*)smem;         // KB*128 (row stride 256 B, swz)
+  uint16_t* vt_lds = k_lds + KB * 128;       // D*VT_STRIDE
 
   const int bh = blockIdx.y;
   const int h = bh % H;
@@ -61,44 +59,38 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   const int q0 = blockIdx.x * QB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int li = lane & 31;   // A-operand row / C col
-  const int hi = lane >> 5;   // half-wave
-  const int qw = q0 + wave * 32;  // this wave's first q row
+  const int li = lane & 31;
+  const int hi = lane >> 5;
+  const int qw = q0 + wave * 32;
+  const int qi = qw + li;  // THIS lane's q row (swapped layout)
   const float slope = slopes[h];
-  const float inv_keep = p_drop > 0.f ? 1.f / (1.f - p_drop) : 1.f;
+  const uint32_t drop_thr = (uint32_t)(p_drop * 65536.0f + 0.5f);
+  const float inv_keep = drop_thr ? 65536.0f / (65536.0f - (float)drop_thr) : 1.0f;
 
-  // ---- Q fragments in registers: A[i=li][kk = s*16 + 8*hi + e] ----
-  constexpr int KS = D / 16;  // QK^T k-steps
+  constexpr int KS = D / 16;
+  constexpr int DB = D / 32;
+
+  // Q fragments (B operand of the swapped QK^T): lane l holds
+  // Q[qw + l%32][8*(l/32) + e] per k-step — a 16 B row-major load.
   bf16x8 q_frag[KS];
   {
-    const int qi = qw + li;
     const bool ok = qi < T;
 #pragma unroll
-    for (int s = 0; s < KS; ++s) {
-      if (ok) {
-        q_frag[s] = *reinterpret_cast<const bf16x8*>(&q[base + (long)qi * D + s * 16 + 8 * hi]);
-      } else {
-        q_frag[s] = bf16x8{};
-      }
-    }
+    for (int s = 0; s < KS; ++s)
+      q_frag[s] = ok ? *reinterpret_cast<const bf16x8*>(
+                           &q[base + (long)qi * D + s * 16 + 8 * hi])
+                     : bf16x8{};
   }
 
-  float m_run[16], l_run[16];
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    m_run[r] = NEG_INF;
-    l_run[r] = 0.f;
-  }
-  constexpr int DB = D / 32;  // PV output column blocks
+  float m_run = NEG_INF, l_run = 0.f;
   f32x16 o_acc[DB];
 #pragma unroll
   for (int d = 0; d < DB; ++d) o_acc[d] = f32x16{};
 
   const int kv_end = min(T, q0 + QB);
-
   for (int kt = 0; kt < kv_end; kt += KB) {
-    // ---- stage K row-major (swizzled) + V transposed ----
-    for (int idx = threadIdx.x * 8; idx < KB * D; idx += 256 * 8) {
+    // ---- stage K row-major (swizzled) + V transposed, all 8 waves ----
+    for (int idx = threadIdx.x * 8; idx < KB * D; idx += 512 * 8) {
       const int key = idx / D, d = idx % D;
       const int kg = kt + key;
       s16x8 kv8{}, vv8{};
@@ -112,74 +104,131 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
     }
     __syncthreads();
 
-    // ---- S = Q @ K^T (32x32 fp32 acc) ----
-    f32x16 s_acc{};
 #pragma unroll
-    for (int s = 0; s < KS; ++s) {
-      const int kk = s * 16 + 8 * hi;
-      bf16x8 k_frag = *reinterpret_cast<const bf16x8*>(
-          (char*)k_lds + swz(li, li * 256 + kk * 2));
-      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q_frag[s], k_frag, s_acc, 0, 0, 0);
-    }
+    for (int sub = 0; sub < 2; ++sub) {
+      const int kt32 = kt + sub * 32;
+      if (kt32 > qw + 31 || kt32 >= T) continue;  // fully masked for this wave
 
-    // ---- online softmax per accumulator register (= per q row) ----
-    const int kj = kt + li;  // this lane's key column
+      // ---- S^T tile: C[i=key][j=q] = mfma(A=K, B=Q^T) ----
+      f32x16 s_acc{};
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      const int qi = qw + row;
-      float sv = s_acc[r] * scale + slope * (float)(kj - qi);
-      if (kj > qi || kj >= T || qi >= T) sv = NEG_INF;
-      const float mr = half_reduce_max(sv);
-      float p = 0.f, alpha = 1.f;
-      if (mr > NEG_INF * 0.5f) {
-        const float mn = fmaxf(m_run[r], mr);
-        alpha = expf(m_run[r] - mn);  // exp(-inf - mn) = 0 on first tile
-        m_run[r] = mn;
-        p = expf(sv - mn);
-      }
-      l_run[r] = l_run[r] * alpha + half_reduce_sum(p);
-#pragma unroll
-      for (int d = 0; d < DB; ++d) o_acc[d][r] *= alpha;
-      // dropout applies to the PV path only (scaled-mask, denominator keeps
-      // the full softmax — dropout acts on normalized probs, reference
-      // layers.py:174)
-      float p_pv = p;
-      if (p_drop > 0.f) {
-        const uint64_t idx = ((uint64_t)bh * (uint64_t)T + (uint64_t)qi) * (uint64_t)T + (uint64_t)kj;
-        p_pv = (uniform01(seed, idx) >= p_drop) ? p * inv_keep : 0.f;
-      }
-      p_lds[(wave * 32 + row) * VT_STRIDE + li] = f32_to_bf16(p_pv);
-    }
-
-    // ---- O += P @ V ----
-#pragma unroll
-    for (int d = 0; d < DB; ++d) {
-#pragma unroll
-      for (int s2 = 0; s2 < KB / 16; ++s2) {
-        const int kk = s2 * 16 + 8 * hi;
+      for (int s = 0; s < KS; ++s) {
+        const int kk = s * 16 + 8 * hi;
+        const int krow = sub * 32 + li;
         bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
-            &p_lds[(wave * 32 + li) * VT_STRIDE + kk]);
-        bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
-            &vt_lds[(d * 32 + li) * VT_STRIDE + kk]);
-        o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_frag, b_frag, o_acc[d], 0, 0, 0);
+            (char*)k_lds + swz(krow, krow * 256 + kk * 2));
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_frag, q_frag[s], s_acc, 0, 0, 0);
+      }
+
+      // ---- in-lane softmax for q row `qi`; reg r holds key kt32+crow(r,hi) ----
+      float sv[16];
+      float tile_max = NEG_INF;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kj = kt32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float x = s_acc[r] * scale + slope * (float)(kj - qi);
+        if (kj > qi || kj >= T || qi >= T) x = NEG_INF;
+        sv[r] = x;
+        tile_max = fmaxf(tile_max, x);
+      }
+      tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
+
+      float alpha = 1.f;
+      const bool valid = tile_max > 0.5f * NEG_INF;
+      if (!__all(tile_max <= m_run)) {  // defer-max, exact threshold 0
+        const float mn = valid ? fmaxf(m_run, tile_max) : m_run;
+        // alpha = exp(old_m - new_m); 0 when old_m was -inf (O, l still zero)
+        alpha = (m_run > 0.5f * NEG_INF) ? __expf(m_run - mn) : (valid ? 0.f : 1.f);
+        m_run = mn;
+      }
+      float p[16];
+      float row_sum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        p[r] = (valid && m_run > 0.5f * NEG_INF) ? __expf(sv[r] - m_run) : 0.f;
+        row_sum += p[r];
+      }
+      row_sum += __shfl_xor(row_sum, 32, 64);
+      l_run = l_run * alpha + row_sum;
+
+      // ---- dropout on the PV path (denominator keeps the full softmax) ----
+      if (drop_thr) {
+        const int bhT_qi = bh * T + qi;
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          const int kbase = kt32 + 8 * g + 4 * hi;  // keys kbase..kbase+3 = regs 4g..4g+3
+          const uint64_t bits = drop_bits(seed, bhT_qi, kbase >> 2);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const bool keep = (uint16_t)(bits >> (16 * e)) >= drop_thr;
+            p[4 * g + e] = keep ? p[4 * g + e] * inv_keep : 0.f;
+          }
+        }
+      }
+
+      // ---- rescale O by alpha of each reg's q row (broadcast via shfl) ----
+      if (!__all(alpha == 1.f)) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float ar = __shfl(alpha, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
+#pragma unroll
+          for (int d = 0; d < DB; ++d) o_acc[d][r] *= ar;
+        }
+      }
+
+      // ---- P -> bf16 A-fragments in-register (cvt_pk + permlane32_swap) ----
+      // words w[j]: cvt_pk(p[2j], p[2j+1]); swap pairs (w0,w2),(w1,w3) and
+      // (w4,w6),(w5,w7) -> lane-correct A[i=q][k=key] fragments.
+      unsigned w[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(w[j]) : "v"(p[2 * j]), "v"(p[2 * j + 1]));
+      }
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        auto r0 = __builtin_amdgcn_permlane32_swap(w[4 * j + 0], w[4 * j + 2], false, false);
+        w[4 * j + 0] = r0[0];
+        w[4 * j + 2] = r0[1];
+        auto r1 = __builtin_amdgcn_permlane32_swap(w[4 * j + 1], w[4 * j + 3], false, false);
+        w[4 * j + 1] = r1[0];
+        w[4 * j + 3] = r1[1];
+      }
+      bf16x8 pa[2];
+      {
+        union { unsigned u[4]; bf16x8 v8; } cvt;
+        cvt.u[0] = w[0]; cvt.u[1] = w[1]; cvt.u[2] = w[2]; cvt.u[3] = w[3];
+        pa[0] = cvt.v8;
+        cvt.u[0] = w[4]; cvt.u[1] = w[5]; cvt.u[2] = w[6]; cvt.u[3] = w[7];
+        pa[1] = cvt.v8;
+      }
+
+      // ---- O += P @ V ----
+#pragma unroll
+      for (int d = 0; d < DB; ++d) {
+#pragma unroll
+        for (int s2 = 0; s2 < 2; ++s2) {
+          bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+              &vt_lds[(d * 32 + li) * VT_STRIDE + sub * 32 + s2 * 16 + 8 * hi]);
+          o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[s2], b_frag, o_acc[d], 0, 0, 0);
+        }
       }
     }
     __syncthreads();
   }
 
-  // ---- epilogue: O / l, lse = m + log(l) ----
+  // ---- epilogue: O rows are crow(r,hi); l/m live in the row's lane ----
+  const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
+  if (hi == 0 && qi < T)
+    lse[(long)bh * T + qi] = l_run > 0.f ? m_run + __logf(l_run) : NEG_INF;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
-    const int qi = qw + row;
-    if (qi >= T) continue;
-    const float inv_l = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+    const int qr = qw + row;
+    if (qr >= T) continue;
+    const float il = __shfl(inv_l, row, 64);
 #pragma unroll
     for (int d = 0; d < DB; ++d)
-      o[base + (long)qi * D + d * 32 + li] = f32_to_bf16(o_acc[d][r] * inv_l);
-    if (li == 0)
-      lse[(long)bh * T + qi] = l_run[r] > 0.f ? m_run[r] + logf(l_run[r]) : NEG_INF;
+      o[base + (long)qr * D + d * 32 + li] = f32_to_bf16(o_acc[d][r] * il);
   }
 }
 
@@ -189,8 +238,8 @@ void launch_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
                 int H, int T, float scale, float p_drop, uint64_t seed,
                 hipStream_t stream) {
   dim3 grid((T + QB - 1) / QB, B * H);
-  const size_t smem = (KB * 128 + D * VT_STRIDE + 4 * 32 * VT_STRIDE) * sizeof(uint16_t);
-  hipLaunchKernelGGL(flash_fwd_kernel<D>, grid, dim3(256), smem, stream,
+  const size_t smem = (KB * 128 + D * VT_STRIDE) * sizeof(uint16_t);
+  hipLaunchKernelGGL(flash_fwd_kernel<D>, grid, dim3(512), smem, stream,
                      (const uint16_t*)q.data_ptr(), (const uint16_t*)k.data_ptr(),
                      (const uint16_t*)v.data_ptr(), slopes.data_ptr<float>(),
                      (uint16_t*)o.data_ptr(), lse.data_ptr<float>(), H, T, scale,

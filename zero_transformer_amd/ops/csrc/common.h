@@ -119,6 +119,14 @@ ZTA_DEV float uniform01(uint64_t seed, uint64_t idx) {
   return float(mix64(seed ^ mix64(idx)) >> 40) * (1.0f / 16777216.0f);
 }
 
+// Attention-dropout bits: 4x 16-bit lanes per hash, shared by forward and
+// backward so masks regenerate identically. Key kj uses hash index
+// (bh*T + qi, kj>>2) and 16-bit lane kj&3; keep iff lane >= threshold
+// (threshold = round(p*65536); rescale by 65536/(65536-threshold)).
+ZTA_DEV uint64_t drop_bits(uint64_t seed, int bhT_qi, int kgroup) {
+  return mix64(seed ^ (((uint64_t)(uint32_t)bhT_qi << 22) | (uint32_t)kgroup));
+}
+
 // ---------------------------------------------------------------------------
 // Grid sizing helper (G11): cap memory-bound grids, grid-stride the rest.
 // ---------------------------------------------------------------------------
