@@ -2,7 +2,11 @@
 shape. Causal flops counted as 2*2*B*H*(T^2/2)*D per GEMM pair."""
 
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -4,17 +4,24 @@
 // re-materializes the dense probs; SURVEY.md §2.3 "Backward of all above").
 //
 //   delta[b,h,i] = sum_d dO * O                  (delta kernel)
-//   dQ kernel : per 128-row Q block, loop KV tiles:
-//       S = Q K^T, P = exp(S*sc + bias - lse), dP = dO V^T,
-//       dS = sc * P (M∘dP/keep - delta), dQ += dS K
-//   dKdV kernel: per 128-key KV block, loop Q tiles:
-//       S^T = K Q^T, P^T, dP^T = V dO^T,
-//       dV += (M∘P^T/keep) dO, dK += sc * (P^T(M∘dP^T/keep - delta)) Q
+//   dQ kernel : per 256-row Q block (8 waves x 32 rows), loop 64-key tiles:
+//       S^T = K Q^T, dP^T = V dO^T   (swapped MFMA: C[i=key][j=q], so each
+//       lane owns one q row — lse/delta are lane scalars),
+//       dS^T = sc * P^T (M∘dP^T/keep - delta),
+//       dQ += dS K  with dS^T -> A-fragments fully in-register
+//       (v_cvt_pk_bf16_f32 + v_permlane32_swap, same transform as the fwd)
+//       and K^T B-fragments via ds_read_b64_tr_b16 hardware-transpose reads
+//       of the row-major XOR-swizzled K tile (no transposed staging copy).
+//   dKdV kernel: per 256-key KV block (8 waves x 32 keys, K/V resident in
+//       registers), loop 64-row Q tiles:
+//       S = Q K^T, dP = dO V^T       (C[i=q][j=key]: key is lane-resident),
+//       dV += P^T dO, dK += sc*(dS^T) Q  with P^T / dS^T A-fragments
+//       in-register and Q^T / dO^T B-fragments via tr-reads.
 //
-// The dropout mask M is regenerated from the counter RNG with the forward's
-// seed. All softmax math in fp32; MFMA operand staging mirrors the forward
-// (row-major XOR-swizzled tiles for k-contiguous B reads, explicitly
-// transposed tiles where the reduction runs over rows).
+// The dropout mask M regenerates from the counter RNG (common.h
+// drop_bits32) with the forward's seed. All softmax math fp32. Staging is
+// T14 async-split: next tile's global loads issue before the compute phase,
+// LDS writes land after the barrier.
 
 #include "common.h"
 
@@ -26,11 +33,12 @@
 namespace {
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(2))) int i32x2;
 
-constexpr int QB = 128;
-constexpr int KB = 32;
+constexpr int NW = 8;         // waves per block
+constexpr int RB = NW * 32;   // rows (q or kv) owned per block
+constexpr int TB = 64;        // staged tile rows per iteration
 constexpr float NEG_INF = -3.0e38f;
-constexpr int TSTRIDE = KB + 8;  // transposed-tile row stride (elements)
 
 ZTA_DEV int swz(int row, int byte_off) { return byte_off ^ ((row & 7) << 4); }
 
@@ -51,39 +59,128 @@ __global__ void delta_kernel(const uint16_t* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// dQ kernel: 4 waves x 32 q rows.
-// LDS: k_lds [KB][128] swz | kt_lds [D][TSTRIDE] | v_lds [KB][128] swz |
-//      ds_lds 4x[32][TSTRIDE]
+// Shared helpers (8-wave kernels, D templated).
+// ---------------------------------------------------------------------------
+
+// B-fragment of mfma_f32_32x32x16_bf16 via hardware transpose read: lane l
+// receives tile[k0 + 8*(l>>5) + e][j0 + (l&31)] for e = 0..7 from a
+// row-major bf16 LDS tile with 256 B row stride and the T2 XOR swizzle.
+//
+// ds_read_b64_tr_b16 semantics (measured, tools/probes/tr_probe.hip): within
+// each 16-lane group, out[lane 4a+b][reg j] = in[lane 4j+a][elem b] — i.e.
+// lane l supplies row ((l>>2)&3), column-block 4*(l&3) of a [4][16] tile and
+// receives the column (l%16) of that tile, rows ascending over the 4 regs.
+ZTA_DEV bf16x8 tr_frag(const uint16_t* lds, int k0, int j0) {
+  const int l = threadIdx.x & 63;
+  const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
+  union {
+    i32x2 d[2];
+    bf16x8 v;
+  } u;
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    const int row = k0 + 8 * (l >> 5) + 4 * half + ((l >> 2) & 3);
+    int addr = (int)(size_t)((const char*)lds + row * 256 +
+                             (colb ^ ((row & 7) << 4)));
+    asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(u.d[half]) : "v"(addr));
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  return u.v;
+}
+
+// In-register C-layout -> A-fragment transform (identical to the forward's
+// T12 path): 16 f32 values x[r] laid out C[i = crow(r,hi)][j = lane&31]
+// become two bf16x8 A-fragments a[s2] with lane l holding
+// A[i = l&31][k = s2*16 + 8*(l>>5) + e].
+ZTA_DEV void c_to_a_frags(const float* x, bf16x8* pa) {
+  unsigned w[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(w[j]) : "v"(x[2 * j]), "v"(x[2 * j + 1]));
+  }
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    auto r0 = __builtin_amdgcn_permlane32_swap(w[4 * j + 0], w[4 * j + 2], false, false);
+    w[4 * j + 0] = r0[0];
+    w[4 * j + 2] = r0[1];
+    auto r1 = __builtin_amdgcn_permlane32_swap(w[4 * j + 1], w[4 * j + 3], false, false);
+    w[4 * j + 1] = r1[0];
+    w[4 * j + 3] = r1[1];
+  }
+  union {
+    unsigned u[4];
+    bf16x8 v8;
+  } cvt;
+  cvt.u[0] = w[0]; cvt.u[1] = w[1]; cvt.u[2] = w[2]; cvt.u[3] = w[3];
+  pa[0] = cvt.v8;
+  cvt.u[0] = w[4]; cvt.u[1] = w[5]; cvt.u[2] = w[6]; cvt.u[3] = w[7];
+  pa[1] = cvt.v8;
+}
+
+// T14 tile staging of a 64-row x D-col bf16 tile into a 256 B-stride
+// swizzled LDS image: issue global loads into registers early (hide HBM
+// latency under the previous tile's compute), write to LDS after the
+// barrier. Thread t owns elements {t*8 + c*4096 | c}, row = idx/D.
 template <int D>
-__global__ __launch_bounds__(256) void flash_dq_kernel(
+struct Stage {
+  static constexpr int NC = (TB * D + 512 * 8 - 1) / (512 * 8);
+  s16x8 r[NC];
+  ZTA_DEV void load(const uint16_t* g, long base, int row0, int T) {
+    const int t = threadIdx.x;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      const int idx = t * 8 + c * 512 * 8;
+      const int rg = row0 + idx / D;
+      r[c] = (idx < TB * D && rg < T)
+                 ? *reinterpret_cast<const s16x8*>(&g[base + (long)rg * D + idx % D])
+                 : s16x8{};
+    }
+  }
+  ZTA_DEV void store(uint16_t* lds) {
+    const int t = threadIdx.x;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      const int idx = t * 8 + c * 512 * 8;
+      if (idx >= TB * D) break;
+      const int row = idx / D, d = idx % D;
+      *reinterpret_cast<s16x8*>((char*)lds + swz(row, row * 256 + d * 2)) = r[c];
+    }
+  }
+};
+
+// ---------------------------------------------------------------------------
+// dQ kernel: 8 waves x 32 q rows = 256-row Q block; loop 64-key KV tiles.
+// LDS: k_lds 64x[256B] swizzled | v_lds 64x[256B] swizzled   (32 KiB)
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(512, 2) void flash_dq_kernel(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     const float* __restrict__ slopes, uint16_t* __restrict__ dq, int H, int T,
-    float scale, float p_drop, uint64_t seed) {
+    float scale, float p_drop, uint32_t seed) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint16_t* k_lds = (uint16_t*)smem;
-  uint16_t* kt_lds = k_lds + KB * 128;
-  uint16_t* v_lds = kt_lds + D * TSTRIDE;
-  uint16_t* ds_lds = v_lds + KB * 128;
+  uint16_t* v_lds = k_lds + TB * 128;
 
   const int bh = blockIdx.y;
   const int h = bh % H;
   const long base = (long)bh * T * D;
-  const int q0 = blockIdx.x * QB;
+  const int q0 = blockIdx.x * RB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int li = lane & 31, hi = lane >> 5;
   const int qw = q0 + wave * 32;
+  const int qi = qw + li;  // this lane's q row
   const float slope = slopes[h];
-  const uint32_t drop_thr = (uint32_t)(p_drop * 65536.0f + 0.5f);
-  const float inv_keep = drop_thr ? 65536.0f / (65536.0f - (float)drop_thr) : 1.0f;
+  const uint32_t thr = (uint32_t)(p_drop * 256.0f + 0.5f);
+  const float inv_keep = thr ? 256.0f / (256.0f - (float)thr) : 1.0f;
 
   constexpr int KS = D / 16;
   constexpr int DB = D / 32;
+
   bf16x8 q_frag[KS], do_frag[KS];
   {
-    const int qi = qw + li;
     const bool ok = qi < T;
 #pragma unroll
     for (int s = 0; s < KS; ++s) {
@@ -96,131 +193,146 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
       }
     }
   }
-  // per-register row constants
-  float lse_r[16], delta_r[16];
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int qi = qw + (r & 3) + 8 * (r >> 2) + 4 * hi;
-    lse_r[r] = qi < T ? lse[(long)bh * T + qi] : 0.f;
-    delta_r[r] = qi < T ? delta[(long)bh * T + qi] : 0.f;
-  }
+  const float lse_q = qi < T ? lse[(long)bh * T + qi] : 0.f;
+  const float delta_q = qi < T ? delta[(long)bh * T + qi] : 0.f;
 
   f32x16 dq_acc[DB];
 #pragma unroll
   for (int d = 0; d < DB; ++d) dq_acc[d] = f32x16{};
 
-  const int kv_end = min(T, q0 + QB);
-  for (int kt = 0; kt < kv_end; kt += KB) {
-    for (int idx = threadIdx.x * 8; idx < KB * D; idx += 256 * 8) {
-      const int key = idx / D, d = idx % D;
-      const int kg = kt + key;
-      s16x8 kv8{}, vv8{};
-      if (kg < T) {
-        kv8 = *reinterpret_cast<const s16x8*>(&k[base + (long)kg * D + d]);
-        vv8 = *reinterpret_cast<const s16x8*>(&v[base + (long)kg * D + d]);
-      }
-      *reinterpret_cast<s16x8*>((char*)k_lds + swz(key, key * 256 + d * 2)) = kv8;
-      *reinterpret_cast<s16x8*>((char*)v_lds + swz(key, key * 256 + d * 2)) = vv8;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) kt_lds[(d + e) * TSTRIDE + key] = (uint16_t)kv8[e];
-    }
-    __syncthreads();
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // static priority for the younger half (T5)
 
-    f32x16 s_acc{}, dp_acc{};
-#pragma unroll
-    for (int s = 0; s < KS; ++s) {
-      const int kk = s * 16 + 8 * hi;
-      bf16x8 k_frag = *reinterpret_cast<const bf16x8*>((char*)k_lds + swz(li, li * 256 + kk * 2));
-      bf16x8 v_frag = *reinterpret_cast<const bf16x8*>((char*)v_lds + swz(li, li * 256 + kk * 2));
-      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q_frag[s], k_frag, s_acc, 0, 0, 0);
-      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(do_frag[s], v_frag, dp_acc, 0, 0, 0);
-    }
-
-    const int kj = kt + li;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      const int qi = qw + row;
-      float p = 0.f;
-      if (!(kj > qi || kj >= T || qi >= T))
-        p = __expf(s_acc[r] * scale + slope * (float)(kj - qi) - lse_r[r]);
-      float dp = dp_acc[r];
-      if (drop_thr) {
-        const uint64_t bits = drop_bits(seed, bh * T + qi, kj >> 2);
-        const bool keep = (uint16_t)(bits >> (16 * (kj & 3))) >= drop_thr;
-        dp = keep ? dp * inv_keep : 0.f;
-      }
-      const float ds = scale * p * (dp - delta_r[r]);
-      ds_lds[(wave * 32 + row) * TSTRIDE + li] = f32_to_bf16(ds);
+  Stage<D> sk, sv;
+  sk.load(k, base, 0, T);
+  sv.load(v, base, 0, T);
+  const int kv_end = min(T, q0 + RB);
+  for (int kt = 0; kt < kv_end; kt += TB) {
+    __syncthreads();  // previous tile fully consumed
+    sk.store(k_lds);
+    sv.store(v_lds);
+    __syncthreads();  // tile kt visible
+    if (kt + TB < kv_end) {
+      sk.load(k, base, kt + TB, T);
+      sv.load(v, base, kt + TB, T);
     }
 
 #pragma unroll
-    for (int d = 0; d < DB; ++d) {
+    for (int sub = 0; sub < 2; ++sub) {
+      const int kt32 = kt + sub * 32;
+      if (kt32 > qw + 31 || kt32 >= T) continue;  // fully masked for this wave
+
+      // S^T = K Q^T ; dP^T = V dO^T   (C[i=key][j=q], q = lane's row)
+      f32x16 s_acc{}, dpt_acc{};
 #pragma unroll
-      for (int s2 = 0; s2 < KB / 16; ++s2) {
-        const int kk = s2 * 16 + 8 * hi;
-        bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(&ds_lds[(wave * 32 + li) * TSTRIDE + kk]);
-        bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(&kt_lds[(d * 32 + li) * TSTRIDE + kk]);
-        dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_frag, b_frag, dq_acc[d], 0, 0, 0);
+      for (int s = 0; s < KS; ++s) {
+        const int kk = s * 16 + 8 * hi;
+        const int krow = sub * 32 + li;
+        bf16x8 ka = *reinterpret_cast<const bf16x8*>((char*)k_lds + swz(krow, krow * 256 + kk * 2));
+        bf16x8 va = *reinterpret_cast<const bf16x8*>((char*)v_lds + swz(krow, krow * 256 + kk * 2));
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, q_frag[s], s_acc, 0, 0, 0);
+        dpt_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, do_frag[s], dpt_acc, 0, 0, 0);
+      }
+
+      // dS^T[key][q] per register (key = kt32 + crow(r,hi), q = qi)
+      float ds[16];
+      if (thr) {
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          const int kbase = kt32 + 8 * g + 4 * hi;
+          const uint32_t bits = drop_bits32(seed, bh * T + qi, kbase >> 2);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const int r = 4 * g + e;
+            const int kj = kbase + e;
+            float p = 0.f;
+            if (!(kj > qi || kj >= T || qi >= T))
+              p = __expf(s_acc[r] * scale + slope * (float)(kj - qi) - lse_q);
+            const bool keep = ((bits >> (8 * e)) & 0xffu) >= thr;
+            const float dp = keep ? dpt_acc[r] * inv_keep : 0.f;
+            ds[r] = scale * p * (dp - delta_q);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kj = kt32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float p = 0.f;
+          if (!(kj > qi || kj >= T || qi >= T))
+            p = __expf(s_acc[r] * scale + slope * (float)(kj - qi) - lse_q);
+          ds[r] = scale * p * (dpt_acc[r] - delta_q);
+        }
+      }
+
+      bf16x8 dsa[2];
+      c_to_a_frags(ds, dsa);  // A[i=q][k=key]
+
+      // dQ += dS K : B[k=key][j=d] via tr-read of the row-major K tile
+#pragma unroll
+      for (int d = 0; d < DB; ++d) {
+#pragma unroll
+        for (int s2 = 0; s2 < 2; ++s2) {
+          bf16x8 kb = tr_frag(k_lds, sub * 32 + s2 * 16, d * 32);
+          dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[s2], kb, dq_acc[d], 0, 0, 0);
+        }
       }
     }
-    __syncthreads();
   }
 
+  // epilogue: dQ rows crow(r,hi)
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
-    const int qi = qw + row;
-    if (qi >= T) continue;
+    const int qr = qw + row;
+    if (qr >= T) continue;
 #pragma unroll
     for (int d = 0; d < DB; ++d)
-      dq[base + (long)qi * D + d * 32 + li] = f32_to_bf16(dq_acc[d][r]);
+      dq[base + (long)qr * D + d * 32 + li] = f32_to_bf16(dq_acc[d][r]);
   }
 }
 
 // ---------------------------------------------------------------------------
-// dKdV kernel: 4 waves x 32 keys = 128-key block, loop q tiles of 32.
-// LDS: q_lds [32][128] swz | qt_lds [D][TSTRIDE] | do_lds [32][128] swz |
-//      dot_lds [D][TSTRIDE] | pT_lds 4x[32][TSTRIDE] | dsT_lds 4x[32][TSTRIDE]
+// dKdV kernel: 8 waves x 32 keys = 256-key block, K/V resident; loop 64-row
+// Q tiles. LDS: q_lds 64x[256B] swz | do_lds 64x[256B] swz | lse/delta 64 f32
+// ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(256) void flash_dkdv_kernel(
+__global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     const float* __restrict__ slopes, uint16_t* __restrict__ dk,
     uint16_t* __restrict__ dv, int H, int T, float scale, float p_drop,
-    uint64_t seed) {
+    uint32_t seed) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint16_t* q_lds = (uint16_t*)smem;
-  uint16_t* qt_lds = q_lds + 32 * 128;
-  uint16_t* do_lds = qt_lds + D * TSTRIDE;
-  uint16_t* dot_lds = do_lds + 32 * 128;
-  uint16_t* pT_lds = dot_lds + D * TSTRIDE;
-  uint16_t* dsT_lds = pT_lds + 4 * 32 * TSTRIDE;
+  uint16_t* do_lds = q_lds + TB * 128;
+  float* lse_s = (float*)(do_lds + TB * 128);
+  float* delta_s = lse_s + TB;
 
   const int bh = blockIdx.y;
   const int h = bh % H;
   const long base = (long)bh * T * D;
-  const int k0 = blockIdx.x * QB;  // 128 keys per block
+  const int k0 = blockIdx.x * RB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int li = lane & 31, hi = lane >> 5;
   const int kw = k0 + wave * 32;  // this wave's first key
+  const int kj = kw + li;         // this lane's key column
   const float slope = slopes[h];
-  const uint32_t drop_thr = (uint32_t)(p_drop * 65536.0f + 0.5f);
-  const float inv_keep = drop_thr ? 65536.0f / (65536.0f - (float)drop_thr) : 1.0f;
+  const uint32_t thr = (uint32_t)(p_drop * 256.0f + 0.5f);
+  const float inv_keep = thr ? 256.0f / (256.0f - (float)thr) : 1.0f;
 
   constexpr int KS = D / 16;
   constexpr int DB = D / 32;
+
   bf16x8 k_frag[KS], v_frag[KS];
   {
-    const int kg = kw + li;
-    const bool ok = kg < T;
+    const bool ok = kj < T;
 #pragma unroll
     for (int s = 0; s < KS; ++s) {
       if (ok) {
-        k_frag[s] = *reinterpret_cast<const bf16x8*>(&k[base + (long)kg * D + s * 16 + 8 * hi]);
-        v_frag[s] = *reinterpret_cast<const bf16x8*>(&v[base + (long)kg * D + s * 16 + 8 * hi]);
+        k_frag[s] = *reinterpret_cast<const bf16x8*>(&k[base + (long)kj * D + s * 16 + 8 * hi]);
+        v_frag[s] = *reinterpret_cast<const bf16x8*>(&v[base + (long)kj * D + s * 16 + 8 * hi]);
       } else {
         k_frag[s] = bf16x8{};
         v_frag[s] = bf16x8{};
@@ -235,86 +347,106 @@ __global__ __launch_bounds__(256) void flash_dkdv_kernel(
     dv_acc[d] = f32x16{};
   }
 
-  for (int qt = k0; qt < T; qt += 32) {
-    // stage Q/dO tiles (row-major swizzled + transposed)
-    for (int idx = threadIdx.x * 8; idx < 32 * D; idx += 256 * 8) {
-      const int qr = idx / D, d = idx % D;
-      const int qg = qt + qr;
-      s16x8 qv8{}, dov8{};
-      if (qg < T) {
-        qv8 = *reinterpret_cast<const s16x8*>(&q[base + (long)qg * D + d]);
-        dov8 = *reinterpret_cast<const s16x8*>(&dout[base + (long)qg * D + d]);
-      }
-      *reinterpret_cast<s16x8*>((char*)q_lds + swz(qr, qr * 256 + d * 2)) = qv8;
-      *reinterpret_cast<s16x8*>((char*)do_lds + swz(qr, qr * 256 + d * 2)) = dov8;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        qt_lds[(d + e) * TSTRIDE + qr] = (uint16_t)qv8[e];
-        dot_lds[(d + e) * TSTRIDE + qr] = (uint16_t)dov8[e];
-      }
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // static priority for the younger half (T5)
+
+  Stage<D> sq, sdo;
+  float lse_reg = 0.f, delta_reg = 0.f;
+  const int t512 = threadIdx.x;
+  auto load_stats = [&](int qt) {
+    if (t512 < TB) {
+      const int qg = qt + t512;
+      lse_reg = qg < T ? lse[(long)bh * T + qg] : 0.f;
+      delta_reg = qg < T ? delta[(long)bh * T + qg] : 0.f;
+    }
+  };
+
+  sq.load(q, base, k0, T);
+  sdo.load(dout, base, k0, T);
+  load_stats(k0);
+  for (int qt = k0; qt < T; qt += TB) {
+    __syncthreads();
+    sq.store(q_lds);
+    sdo.store(do_lds);
+    if (t512 < TB) {
+      lse_s[t512] = lse_reg;
+      delta_s[t512] = delta_reg;
     }
     __syncthreads();
-
-    // S^T = K Q^T ; dP^T = V dO^T
-    f32x16 st_acc{}, dpt_acc{};
-#pragma unroll
-    for (int s = 0; s < KS; ++s) {
-      const int kk = s * 16 + 8 * hi;
-      bf16x8 qb = *reinterpret_cast<const bf16x8*>((char*)q_lds + swz(li, li * 256 + kk * 2));
-      bf16x8 dob = *reinterpret_cast<const bf16x8*>((char*)do_lds + swz(li, li * 256 + kk * 2));
-      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(k_frag[s], qb, st_acc, 0, 0, 0);
-      dpt_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(v_frag[s], dob, dpt_acc, 0, 0, 0);
+    if (qt + TB < T) {
+      sq.load(q, base, qt + TB, T);
+      sdo.load(dout, base, qt + TB, T);
+      load_stats(qt + TB);
     }
 
-    const int qi = qt + li;  // this lane's q column
-    const float lse_q = qi < T ? lse[(long)bh * T + qi] : 0.f;
-    const float delta_q = qi < T ? delta[(long)bh * T + qi] : 0.f;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      const int kj = kw + row;
-      float p = 0.f;
-      if (!(kj > qi || kj >= T || qi >= T))
-        p = __expf(st_acc[r] * scale + slope * (float)(kj - qi) - lse_q);
-      float dp = dpt_acc[r];
-      float p_pv = p;
-      if (drop_thr) {
-        const uint64_t bits = drop_bits(seed, bh * T + qi, kj >> 2);
-        const bool keep = (uint16_t)(bits >> (16 * (kj & 3))) >= drop_thr;
-        dp = keep ? dp * inv_keep : 0.f;
-        p_pv = keep ? p * inv_keep : 0.f;
+    for (int sub = 0; sub < 2; ++sub) {
+      const int q32 = qt + sub * 32;
+      if (q32 + 31 < kw || q32 >= T) continue;  // q < key: fully masked
+
+      // S = Q K^T ; dP = dO V^T   (C[i=q][j=key], key = lane's column)
+      f32x16 s_acc{}, dp_acc{};
+#pragma unroll
+      for (int s = 0; s < KS; ++s) {
+        const int kk = s * 16 + 8 * hi;
+        const int qrow = sub * 32 + li;
+        bf16x8 qa = *reinterpret_cast<const bf16x8*>((char*)q_lds + swz(qrow, qrow * 256 + kk * 2));
+        bf16x8 da = *reinterpret_cast<const bf16x8*>((char*)do_lds + swz(qrow, qrow * 256 + kk * 2));
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, k_frag[s], s_acc, 0, 0, 0);
+        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, v_frag[s], dp_acc, 0, 0, 0);
       }
-      const float ds = scale * p * (dp - delta_q);
-      dsT_lds[(wave * 32 + row) * TSTRIDE + li] = f32_to_bf16(ds);
-      pT_lds[(wave * 32 + row) * TSTRIDE + li] = f32_to_bf16(p_pv);
-    }
 
-    // dK += dS^T Q ; dV += P^T dO   (reduction over the 32 q columns)
+      // P^T (for dV) and dS^T (for dK) per register; q = q32 + crow(r,hi)
+      float p_pv[16], ds[16];
 #pragma unroll
-    for (int d = 0; d < DB; ++d) {
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const int qi = q32 + row;
+        const float lq = lse_s[sub * 32 + row];
+        const float dq_ = delta_s[sub * 32 + row];
+        float p = 0.f;
+        if (!(kj > qi || kj >= T || qi >= T))
+          p = __expf(s_acc[r] * scale + slope * (float)(kj - qi) - lq);
+        float dp = dp_acc[r];
+        float ppv = p;
+        if (thr) {
+          const uint32_t bits = drop_bits32(seed, bh * T + qi, kj >> 2);
+          const bool keep = ((bits >> (8 * (kj & 3))) & 0xffu) >= thr;
+          dp = keep ? dp * inv_keep : 0.f;
+          ppv = keep ? p * inv_keep : 0.f;
+        }
+        p_pv[r] = ppv;
+        ds[r] = scale * p * (dp - dq_);
+      }
+
+      bf16x8 pa[2], dsa[2];
+      c_to_a_frags(p_pv, pa);  // A[i=key][k=q]
+      c_to_a_frags(ds, dsa);
+
+      // dV += P^T dO ; dK += dS^T Q : B[k=q][j=d] via tr-reads
 #pragma unroll
-      for (int s2 = 0; s2 < 32 / 16; ++s2) {
-        const int kk = s2 * 16 + 8 * hi;
-        bf16x8 dsa = *reinterpret_cast<const bf16x8*>(&dsT_lds[(wave * 32 + li) * TSTRIDE + kk]);
-        bf16x8 pa = *reinterpret_cast<const bf16x8*>(&pT_lds[(wave * 32 + li) * TSTRIDE + kk]);
-        bf16x8 qtb = *reinterpret_cast<const bf16x8*>(&qt_lds[(d * 32 + li) * TSTRIDE + kk]);
-        bf16x8 dotb = *reinterpret_cast<const bf16x8*>(&dot_lds[(d * 32 + li) * TSTRIDE + kk]);
-        dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qtb, dk_acc[d], 0, 0, 0);
-        dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dotb, dv_acc[d], 0, 0, 0);
+      for (int d = 0; d < DB; ++d) {
+#pragma unroll
+        for (int s2 = 0; s2 < 2; ++s2) {
+          bf16x8 dob = tr_frag(do_lds, sub * 32 + s2 * 16, d * 32);
+          bf16x8 qb = tr_frag(q_lds, sub * 32 + s2 * 16, d * 32);
+          dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[s2], dob, dv_acc[d], 0, 0, 0);
+          dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[s2], qb, dk_acc[d], 0, 0, 0);
+        }
       }
     }
-    __syncthreads();
   }
 
+  // epilogue: dK/dV rows crow(r,hi)
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
-    const int kj = kw + row;
-    if (kj >= T) continue;
+    const int kr = kw + row;
+    if (kr >= T) continue;
 #pragma unroll
     for (int d = 0; d < DB; ++d) {
-      dk[base + (long)kj * D + d * 32 + li] = f32_to_bf16(dk_acc[d][r]);
-      dv[base + (long)kj * D + d * 32 + li] = f32_to_bf16(dv_acc[d][r]);
+      dk[base + (long)kr * D + d * 32 + li] = f32_to_bf16(dk_acc[d][r]);
+      dv[base + (long)kr * D + d * 32 + li] = f32_to_bf16(dv_acc[d][r]);
     }
   }
 }
@@ -324,19 +456,17 @@ void launch_bwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
                 const at::Tensor& dout, const at::Tensor& lse, const at::Tensor& delta,
                 const at::Tensor& slopes, at::Tensor& dq, at::Tensor& dk,
                 at::Tensor& dv, int B, int H, int T, float scale, float p_drop,
-                uint64_t seed, hipStream_t stream) {
-  dim3 grid((T + QB - 1) / QB, B * H);
-  const size_t smem_dq =
-      (KB * 128 + D * TSTRIDE + KB * 128 + 4 * 32 * TSTRIDE) * sizeof(uint16_t);
-  hipLaunchKernelGGL(flash_dq_kernel<D>, grid, dim3(256), smem_dq, stream,
+                uint32_t seed, hipStream_t stream) {
+  dim3 grid((T + RB - 1) / RB, B * H);
+  const size_t smem_dq = 2 * TB * 128 * sizeof(uint16_t);
+  hipLaunchKernelGGL(flash_dq_kernel<D>, grid, dim3(512), smem_dq, stream,
                      (const uint16_t*)q.data_ptr(), (const uint16_t*)k.data_ptr(),
                      (const uint16_t*)v.data_ptr(), (const uint16_t*)dout.data_ptr(),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
                      slopes.data_ptr<float>(), (uint16_t*)dq.data_ptr(), H, T, scale,
                      p_drop, seed);
-  const size_t smem_kv =
-      (2 * 32 * 128 + 2 * D * TSTRIDE + 8 * 32 * TSTRIDE) * sizeof(uint16_t);
-  hipLaunchKernelGGL(flash_dkdv_kernel<D>, grid, dim3(256), smem_kv, stream,
+  const size_t smem_kv = 2 * TB * 128 * sizeof(uint16_t) + 2 * TB * sizeof(float);
+  hipLaunchKernelGGL(flash_dkdv_kernel<D>, grid, dim3(512), smem_kv, stream,
                      (const uint16_t*)q.data_ptr(), (const uint16_t*)k.data_ptr(),
                      (const uint16_t*)v.data_ptr(), (const uint16_t*)dout.data_ptr(),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
@@ -366,10 +496,10 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                        (const uint16_t*)o.data_ptr(), delta.data_ptr<float>(), rows, D);
   }
   switch (D) {
-    case 32: launch_bwd<32>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;
-    case 64: launch_bwd<64>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;
-    case 96: launch_bwd<96>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;
-    case 128: launch_bwd<128>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint64_t)seed, stream); break;
+    case 32: launch_bwd<32>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 64: launch_bwd<64>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 96: launch_bwd<96>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 128: launch_bwd<128>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
     default: TORCH_CHECK(false, "attn_bwd: head_dim must be 32/64/96/128, got ", D);
   }
   return {dq, dk, dv};

@@ -64,8 +64,8 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
   const int qw = q0 + wave * 32;
   const int qi = qw + li;  // THIS lane's q row (swapped layout)
   const float slope = slopes[h];
-  const uint32_t drop_thr = (uint32_t)(p_drop * 65536.0f + 0.5f);
-  const float inv_keep = drop_thr ? 65536.0f / (65536.0f - (float)drop_thr) : 1.0f;
+  const uint32_t drop_thr = (uint32_t)(p_drop * 256.0f + 0.5f);
+  const float inv_keep = drop_thr ? 256.0f / (256.0f - (float)drop_thr) : 1.0f;
 
   constexpr int KS = D / 16;
   constexpr int DB = D / 32;
@@ -157,10 +157,10 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
           const int kbase = kt32 + 8 * g + 4 * hi;  // keys kbase..kbase+3 = regs 4g..4g+3
-          const uint64_t bits = drop_bits(seed, bhT_qi, kbase >> 2);
+          const uint32_t bits = drop_bits32((uint32_t)seed, bhT_qi, kbase >> 2);
 #pragma unroll
           for (int e = 0; e < 4; ++e) {
-            const bool keep = (uint16_t)(bits >> (16 * e)) >= drop_thr;
+            const bool keep = ((bits >> (8 * e)) & 0xffu) >= drop_thr;
             p[4 * g + e] = keep ? p[4 * g + e] * inv_keep : 0.f;
           }
         }

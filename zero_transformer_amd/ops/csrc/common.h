@@ -119,12 +119,27 @@ ZTA_DEV float uniform01(uint64_t seed, uint64_t idx) {
   return float(mix64(seed ^ mix64(idx)) >> 40) * (1.0f / 16777216.0f);
 }
 
-// Attention-dropout bits: 4x 16-bit lanes per hash, shared by forward and
-// backward so masks regenerate identically. Key kj uses hash index
-// (bh*T + qi, kj>>2) and 16-bit lane kj&3; keep iff lane >= threshold
-// (threshold = round(p*65536); rescale by 65536/(65536-threshold)).
-ZTA_DEV uint64_t drop_bits(uint64_t seed, int bhT_qi, int kgroup) {
-  return mix64(seed ^ (((uint64_t)(uint32_t)bhT_qi << 22) | (uint32_t)kgroup));
+// ---------------------------------------------------------------------------
+// Attention-dropout RNG: murmur3-finalizer 32-bit hash, 4x 8-bit thresholds
+// per call, shared by the fwd / dq / dkdv kernels so the mask regenerates
+// identically in every orientation (and cheaply: ~8 VALU per 4 elements —
+// the dkdv kernel evaluates it with q varying per register, 16 calls per
+// subtile, so hash cost matters). Element (qi, kj) keeps iff byte
+// (bits(seed, bh*T+qi, kj>>2) >> 8*(kj&3)) & 0xff >= thr, thr = round(p*256);
+// rescale by 256/(256-thr). Mirrored in Python by ops.reference.drop_mask.
+// ---------------------------------------------------------------------------
+ZTA_DEV uint32_t mix32(uint32_t x) {
+  x ^= x >> 16;
+  x *= 0x85ebca6bu;
+  x ^= x >> 13;
+  x *= 0xc2b2ae35u;
+  x ^= x >> 16;
+  return x;
+}
+
+ZTA_DEV uint32_t drop_bits32(uint32_t seed, int bhT_qi, int kgroup) {
+  return mix32(seed ^ ((uint32_t)bhT_qi * 0x9e3779b9u) ^
+               ((uint32_t)kgroup * 0x85ebca6bu));
 }
 
 // ---------------------------------------------------------------------------

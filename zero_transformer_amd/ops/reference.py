@@ -73,6 +73,73 @@ def attention(
     return out.to(q.dtype)
 
 
+def drop_mask(seed: int, B: int, H: int, T: int, p: float) -> torch.Tensor:
+    """Regenerate the attention-dropout keep mask of the HIP kernels.
+
+    Exact Python mirror of csrc/common.h drop_bits32 (murmur3-finalizer
+    mix32; 4x 8-bit thresholds per hash over key groups of 4). Returns a
+    bool (B, H, T, T) tensor: True = keep. Used by GPU parity tests to
+    compare kernel backward against autograd with the explicit mask.
+    """
+    import numpy as np
+
+    thr = int(p * 256.0 + 0.5)
+
+    def mix32(x):
+        x = x.astype(np.uint32, copy=True)
+        x ^= x >> np.uint32(16)
+        x *= np.uint32(0x85EBCA6B)
+        x ^= x >> np.uint32(13)
+        x *= np.uint32(0xC2B2AE35)
+        x ^= x >> np.uint32(16)
+        return x
+
+    bh = np.arange(B * H, dtype=np.uint32)
+    qi = np.arange(T, dtype=np.uint32)
+    kg = np.arange((T + 3) // 4, dtype=np.uint32)
+    bhT_qi = (bh[:, None] * np.uint32(T) + qi[None, :]).reshape(-1, 1)  # (BH*T, 1)
+    with np.errstate(over="ignore"):
+        x = (
+            np.uint32(seed)
+            ^ (bhT_qi * np.uint32(0x9E3779B9))
+            ^ (kg[None, :] * np.uint32(0x85EBCA6B))
+        )
+        bits = mix32(x)  # (BH*T, ceil(T/4))
+    bytes_ = np.stack([(bits >> np.uint32(8 * e)) & np.uint32(0xFF) for e in range(4)], axis=-1)
+    keep = (bytes_ >= np.uint32(thr)).reshape(B * H * T, -1)[:, :T]
+    return torch.from_numpy(keep.reshape(B, H, T, T).copy())
+
+
+def drop_inv_keep(p: float) -> float:
+    """Rescale factor matching the kernels' realized 8-bit threshold."""
+    thr = int(p * 256.0 + 0.5)
+    return 256.0 / (256.0 - thr) if thr else 1.0
+
+
+def attention_with_mask(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    slopes: Optional[torch.Tensor],
+    keep: torch.Tensor,
+    inv_keep: float,
+) -> torch.Tensor:
+    """Reference attention with an explicit dropout keep-mask on the probs
+    (denominator keeps the full softmax, matching the fused kernel)."""
+    B, H, T, D = q.shape
+    scores = (q.float() @ k.float().transpose(-1, -2)) / math.sqrt(D)
+    if slopes is not None:
+        pos = torch.arange(T, device=q.device, dtype=torch.float32)
+        rel = pos.view(1, T) - pos.view(T, 1)
+        scores = scores + slopes.to(q.device).float().view(1, H, 1, 1) * rel.view(1, 1, T, T)
+    causal = torch.ones(T, T, dtype=torch.bool, device=q.device).tril()
+    scores = scores.masked_fill(~causal.view(1, 1, T, T), torch.finfo(torch.float32).min)
+    probs = F.softmax(scores, dim=-1)
+    probs = probs * keep.to(probs.dtype).to(probs.device) * inv_keep
+    out = probs.to(v.dtype) @ v
+    return out.to(q.dtype)
+
+
 def layer_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
     """Bias-free LayerNorm in fp32, cast back to input dtype.
 
