@@ -43,6 +43,7 @@ def _torch_flags():
 
 def _src_hash() -> str:
     h = hashlib.sha256()
+    h.update(os.environ.get("ZTA_HIPCC_EXTRA", "").encode())
     for s in SOURCES + ["common.h"]:
         p = os.path.join(CSRC, s)
         if os.path.exists(p):
@@ -72,6 +73,9 @@ def build(force: bool = False, verbose: bool = True) -> str:
         "-fno-gpu-rdc",
         "-Wno-unused-result",
     ]
+    extra = os.environ.get("ZTA_HIPCC_EXTRA", "")
+    if extra:
+        cmd += extra.split()
     for i in includes:
         cmd += ["-I", i]
     cmd += [os.path.join(CSRC, s) for s in SOURCES]

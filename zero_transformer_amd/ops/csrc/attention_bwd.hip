@@ -23,6 +23,7 @@
 // T14 async-split: next tile's global loads issue before the compute phase,
 // LDS writes land after the barrier.
 
+#include "attn_tiles.h"
 #include "common.h"
 
 #include <ATen/ATen.h>
@@ -32,15 +33,15 @@
 
 namespace {
 
-typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
-typedef __attribute__((ext_vector_type(2))) int i32x2;
+using attn::NW;
+using attn::RB;
+using attn::TB;
+using attn::Stage;
+using attn::c_to_a_frags;
+using attn::swz;
+using attn::tr_frag;
 
-constexpr int NW = 8;         // waves per block
-constexpr int RB = NW * 32;   // rows (q or kv) owned per block
-constexpr int TB = 64;        // staged tile rows per iteration
 constexpr float NEG_INF = -3.0e38f;
-
-ZTA_DEV int swz(int row, int byte_off) { return byte_off ^ ((row & 7) << 4); }
 
 // ---------------------------------------------------------------------------
 __global__ void delta_kernel(const uint16_t* __restrict__ dout,
@@ -59,95 +60,6 @@ __global__ void delta_kernel(const uint16_t* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// Shared helpers (8-wave kernels, D templated).
-// ---------------------------------------------------------------------------
-
-// B-fragment of mfma_f32_32x32x16_bf16 via hardware transpose read: lane l
-// receives tile[k0 + 8*(l>>5) + e][j0 + (l&31)] for e = 0..7 from a
-// row-major bf16 LDS tile with 256 B row stride and the T2 XOR swizzle.
-//
-// ds_read_b64_tr_b16 semantics (measured, tools/probes/tr_probe.hip): within
-// each 16-lane group, out[lane 4a+b][reg j] = in[lane 4j+a][elem b] — i.e.
-// lane l supplies row ((l>>2)&3), column-block 4*(l&3) of a [4][16] tile and
-// receives the column (l%16) of that tile, rows ascending over the 4 regs.
-ZTA_DEV bf16x8 tr_frag(const uint16_t* lds, int k0, int j0) {
-  const int l = threadIdx.x & 63;
-  const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
-  union {
-    i32x2 d[2];
-    bf16x8 v;
-  } u;
-#pragma unroll
-  for (int half = 0; half < 2; ++half) {
-    const int row = k0 + 8 * (l >> 5) + 4 * half + ((l >> 2) & 3);
-    int addr = (int)(size_t)((const char*)lds + row * 256 +
-                             (colb ^ ((row & 7) << 4)));
-    asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(u.d[half]) : "v"(addr));
-  }
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  return u.v;
-}
-
-// In-register C-layout -> A-fragment transform (identical to the forward's
-// T12 path): 16 f32 values x[r] laid out C[i = crow(r,hi)][j = lane&31]
-// become two bf16x8 A-fragments a[s2] with lane l holding
-// A[i = l&31][k = s2*16 + 8*(l>>5) + e].
-ZTA_DEV void c_to_a_frags(const float* x, bf16x8* pa) {
-  unsigned w[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(w[j]) : "v"(x[2 * j]), "v"(x[2 * j + 1]));
-  }
-#pragma unroll
-  for (int j = 0; j < 2; ++j) {
-    auto r0 = __builtin_amdgcn_permlane32_swap(w[4 * j + 0], w[4 * j + 2], false, false);
-    w[4 * j + 0] = r0[0];
-    w[4 * j + 2] = r0[1];
-    auto r1 = __builtin_amdgcn_permlane32_swap(w[4 * j + 1], w[4 * j + 3], false, false);
-    w[4 * j + 1] = r1[0];
-    w[4 * j + 3] = r1[1];
-  }
-  union {
-    unsigned u[4];
-    bf16x8 v8;
-  } cvt;
-  cvt.u[0] = w[0]; cvt.u[1] = w[1]; cvt.u[2] = w[2]; cvt.u[3] = w[3];
-  pa[0] = cvt.v8;
-  cvt.u[0] = w[4]; cvt.u[1] = w[5]; cvt.u[2] = w[6]; cvt.u[3] = w[7];
-  pa[1] = cvt.v8;
-}
-
-// T14 tile staging of a 64-row x D-col bf16 tile into a 256 B-stride
-// swizzled LDS image: issue global loads into registers early (hide HBM
-// latency under the previous tile's compute), write to LDS after the
-// barrier. Thread t owns elements {t*8 + c*4096 | c}, row = idx/D.
-template <int D>
-struct Stage {
-  static constexpr int NC = (TB * D + 512 * 8 - 1) / (512 * 8);
-  s16x8 r[NC];
-  ZTA_DEV void load(const uint16_t* g, long base, int row0, int T) {
-    const int t = threadIdx.x;
-#pragma unroll
-    for (int c = 0; c < NC; ++c) {
-      const int idx = t * 8 + c * 512 * 8;
-      const int rg = row0 + idx / D;
-      r[c] = (idx < TB * D && rg < T)
-                 ? *reinterpret_cast<const s16x8*>(&g[base + (long)rg * D + idx % D])
-                 : s16x8{};
-    }
-  }
-  ZTA_DEV void store(uint16_t* lds) {
-    const int t = threadIdx.x;
-#pragma unroll
-    for (int c = 0; c < NC; ++c) {
-      const int idx = t * 8 + c * 512 * 8;
-      if (idx >= TB * D) break;
-      const int row = idx / D, d = idx % D;
-      *reinterpret_cast<s16x8*>((char*)lds + swz(row, row * 256 + d * 2)) = r[c];
-    }
-  }
-};
-
 // ---------------------------------------------------------------------------
 // dQ kernel: 8 waves x 32 q rows = 256-row Q block; loop 64-key KV tiles.
 // LDS: k_lds 64x[256B] swizzled | v_lds 64x[256B] swizzled   (32 KiB)

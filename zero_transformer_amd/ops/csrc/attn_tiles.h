@@ -1,0 +1,117 @@
+// Shared LDS-tile helpers for the flash attention forward/backward kernels
+// (gfx950). Row-major 64-row x D-col bf16 tiles at a fixed 256 B row stride
+// with the T2 XOR swizzle; MFMA fragment production via hardware transpose
+// reads and in-register cvt_pk+permlane transforms.
+#pragma once
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(2))) int i32x2;
+
+namespace attn {
+
+constexpr int NW = 8;        // waves per block
+constexpr int RB = NW * 32;  // rows (q or kv) owned per block
+constexpr int TB = 64;       // staged tile rows per iteration
+
+ZTA_DEV int swz(int row, int byte_off) { return byte_off ^ ((row & 7) << 4); }
+
+// B-fragment of mfma_f32_32x32x16_bf16 via hardware transpose read: lane l
+// receives tile[k0 + 8*(l>>5) + e][j0 + (l&31)] for e = 0..7 from a
+// row-major bf16 LDS tile with 256 B row stride and the T2 XOR swizzle.
+//
+// ds_read_b64_tr_b16 semantics (measured, tools/probes/tr_probe.hip): within
+// each 16-lane group, out[lane 4a+b][reg j] = in[lane 4j+a][elem b] — i.e.
+// lane l supplies row ((l>>2)&3), column-block 4*(l&3) of a [4][16] tile and
+// receives the column (l%16) of that tile, rows ascending over the 4 regs.
+// NOTE: reads + waitcnt MUST be one asm statement. SIInsertWaitcnts cannot
+// see a ds_read inside inline asm, so it inserts no lgkmcnt wait before uses
+// of its outputs, and a separate waitcnt asm has no dataflow edge to the
+// outputs — the scheduler may move the consuming MFMA between read and wait
+// (observed: register-junk O values with exact lse). Bundling makes every
+// consumer order after the wait by SSA construction; "=&v" keeps the
+// destinations from aliasing the still-live address operands.
+ZTA_DEV bf16x8 tr_frag(const uint16_t* lds, int k0, int j0) {
+  const int l = threadIdx.x & 63;
+  const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
+  const int r0 = k0 + 8 * (l >> 5) + ((l >> 2) & 3);
+  const int a0 = (int)(size_t)((const char*)lds + r0 * 256 + (colb ^ ((r0 & 7) << 4)));
+  const int r1 = r0 + 4;
+  const int a1 = (int)(size_t)((const char*)lds + r1 * 256 + (colb ^ ((r1 & 7) << 4)));
+  union {
+    i32x2 d[2];
+    bf16x8 v;
+  } u;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(u.d[0]), "=&v"(u.d[1])
+      : "v"(a0), "v"(a1)
+      : "memory");  // the asm READS LDS: without the clobber the compiler
+                    // neither orders other waves' staging stores against it
+                    // nor keeps them alive (no visible reader)
+  return u.v;
+}
+
+// In-register C-layout -> A-fragment transform (T12): 16 f32 values x[r]
+// laid out C[i = crow(r,hi)][j = lane&31] become two bf16x8 A-fragments
+// pa[s2] with lane l holding A[i = l&31][k = s2*16 + 8*(l>>5) + e].
+ZTA_DEV void c_to_a_frags(const float* x, bf16x8* pa) {
+  unsigned w[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(w[j]) : "v"(x[2 * j]), "v"(x[2 * j + 1]));
+  }
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    auto r0 = __builtin_amdgcn_permlane32_swap(w[4 * j + 0], w[4 * j + 2], false, false);
+    w[4 * j + 0] = r0[0];
+    w[4 * j + 2] = r0[1];
+    auto r1 = __builtin_amdgcn_permlane32_swap(w[4 * j + 1], w[4 * j + 3], false, false);
+    w[4 * j + 1] = r1[0];
+    w[4 * j + 3] = r1[1];
+  }
+  union {
+    unsigned u[4];
+    bf16x8 v8;
+  } cvt;
+  cvt.u[0] = w[0]; cvt.u[1] = w[1]; cvt.u[2] = w[2]; cvt.u[3] = w[3];
+  pa[0] = cvt.v8;
+  cvt.u[0] = w[4]; cvt.u[1] = w[5]; cvt.u[2] = w[6]; cvt.u[3] = w[7];
+  pa[1] = cvt.v8;
+}
+
+// T14 tile staging of a 64-row x D-col bf16 tile into a 256 B-stride
+// swizzled LDS image: issue global loads into registers early (hide HBM
+// latency under the previous tile's compute), write to LDS after the
+// barrier. Thread t owns elements {t*8 + c*4096 | c}, row = idx/D.
+template <int D>
+struct Stage {
+  static constexpr int NC = (TB * D + 512 * 8 - 1) / (512 * 8);
+  s16x8 r[NC];
+  ZTA_DEV void load(const uint16_t* g, long base, int row0, int T) {
+    const int t = threadIdx.x;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      const int idx = t * 8 + c * 512 * 8;
+      const int rg = row0 + idx / D;
+      r[c] = (idx < TB * D && rg < T)
+                 ? *reinterpret_cast<const s16x8*>(&g[base + (long)rg * D + idx % D])
+                 : s16x8{};
+    }
+  }
+  ZTA_DEV void store(uint16_t* lds) {
+    const int t = threadIdx.x;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      const int idx = t * 8 + c * 512 * 8;
+      if (idx >= TB * D) break;
+      const int row = idx / D, d = idx % D;
+      *reinterpret_cast<s16x8*>((char*)lds + swz(row, row * 256 + d * 2)) = r[c];
+    }
+  }
+};
+
+}  // namespace attn

@@ -1,7 +1,10 @@
 // Bias-free LayerNorm forward + backward for gfx950.
 // Replaces the XLA-fused LayerNorm of the reference (GPT.py:42,49,98 — 49
-// calls/step at 1.3B). Memory-bound: vectorized 8x bf16 loads (G13), one
-// block per row batch, fp32 statistics.
+// calls/step at 1.3B). Memory-bound: the bf16 fast path is fully vectorized
+// (s16x8 loads, G13), keeps the row in registers (one read + one write per
+// tensor), and accumulates dw in per-thread registers across the row loop
+// (columns are thread-owned), with one atomic pass at kernel end.
+// fp32 statistics throughout.
 
 #include "common.h"
 
@@ -12,15 +15,108 @@
 
 namespace {
 
+// ---------------------------------------------------------------------------
+// Fast path: bf16, C % 8 == 0, block = C/8 threads (<= 1024). The whole row
+// lives in one s16x8 per thread.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024) void ln_fwd_vec(
+    const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
+    uint16_t* __restrict__ y, float* __restrict__ mean_out,
+    float* __restrict__ rstd_out, long rows, int C, float eps) {
+  __shared__ float scratch[16];
+  const int t = threadIdx.x;
+  float wv[8];
+  {
+    s16x8 w8 = *reinterpret_cast<const s16x8*>(&w[t * 8]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) wv[e] = bf16_to_f32((uint16_t)w8[e]);
+  }
+  const float invC = 1.0f / (float)C;
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    s16x8 x8 = *reinterpret_cast<const s16x8*>(&x[row * C + t * 8]);
+    float xv[8];
+    float sum = 0.f, sumsq = 0.f;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      xv[e] = bf16_to_f32((uint16_t)x8[e]);
+      sum += xv[e];
+      sumsq += xv[e] * xv[e];
+    }
+    sum = block_reduce_sum(sum, scratch);
+    sumsq = block_reduce_sum(sumsq, scratch);
+    const float mu = sum * invC;
+    const float rstd = rsqrtf(sumsq * invC - mu * mu + eps);
+    if (t == 0) {
+      mean_out[row] = mu;
+      rstd_out[row] = rstd;
+    }
+    s16x8 y8;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      y8[e] = (short)f32_to_bf16((xv[e] - mu) * rstd * wv[e]);
+    *reinterpret_cast<s16x8*>(&y[row * C + t * 8]) = y8;
+  }
+}
+
+// dx = rstd * (g - mean(g) - xhat * mean(g*xhat)),  g = w * dy
+// dw[c] = sum_rows dy[r,c] * xhat[r,c]  — thread-owned columns, register
+// accumulation across the row loop, one atomicAdd per column at the end.
+__global__ __launch_bounds__(1024) void ln_bwd_vec(
+    const uint16_t* __restrict__ dy, const uint16_t* __restrict__ x,
+    const uint16_t* __restrict__ w, const float* __restrict__ mean,
+    const float* __restrict__ rstd, uint16_t* __restrict__ dx,
+    float* __restrict__ dw_f32, long rows, int C) {
+  __shared__ float scratch[16];
+  const int t = threadIdx.x;
+  float wv[8], dw_acc[8];
+  {
+    s16x8 w8 = *reinterpret_cast<const s16x8*>(&w[t * 8]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      wv[e] = bf16_to_f32((uint16_t)w8[e]);
+      dw_acc[e] = 0.f;
+    }
+  }
+  const float invC = 1.0f / (float)C;
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    s16x8 dy8 = *reinterpret_cast<const s16x8*>(&dy[row * C + t * 8]);
+    s16x8 x8 = *reinterpret_cast<const s16x8*>(&x[row * C + t * 8]);
+    const float mu = mean[row], rs = rstd[row];
+    float g[8], xh[8];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float d = bf16_to_f32((uint16_t)dy8[e]);
+      xh[e] = (bf16_to_f32((uint16_t)x8[e]) - mu) * rs;
+      g[e] = d * wv[e];
+      s1 += g[e];
+      s2 += g[e] * xh[e];
+      dw_acc[e] += d * xh[e];
+    }
+    s1 = block_reduce_sum(s1, scratch) * invC;
+    s2 = block_reduce_sum(s2, scratch) * invC;
+    s16x8 dx8;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      dx8[e] = (short)f32_to_bf16((g[e] - s1 - xh[e] * s2) * rs);
+    *reinterpret_cast<s16x8*>(&dx[row * C + t * 8]) = dx8;
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) atomicAdd(&dw_f32[t * 8 + e], dw_acc[e]);
+}
+
+// ---------------------------------------------------------------------------
+// Generic fallback (any dtype / any C)
+// ---------------------------------------------------------------------------
 template <typename T>
 __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
                               T* __restrict__ y, float* __restrict__ mean_out,
-                              float* __restrict__ rstd_out, int rows, int C,
+                              float* __restrict__ rstd_out, long rows, int C,
                               float eps) {
   __shared__ float scratch[16];
-  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
-    const T* xr = x + (long)row * C;
-    T* yr = y + (long)row * C;
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + row * C;
+    T* yr = y + row * C;
     float sum = 0.f, sumsq = 0.f;
     for (int c = threadIdx.x; c < C; c += blockDim.x) {
       float v = to_f32(xr[c]);
@@ -43,25 +139,21 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
   }
 }
 
-// dx = rstd * w * dy  - rstd * mean(w*dy)  - xhat * rstd * mean(w*dy*xhat)
-// dw[c] = sum_rows dy[r,c] * xhat[r,c]   (block-local fp32 accumulation in
-// LDS, one global atomic pass per block — avoids a [blocks, C] partial
-// buffer).
 template <typename T>
 __global__ void ln_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                               const T* __restrict__ w, const float* __restrict__ mean,
                               const float* __restrict__ rstd, T* __restrict__ dx,
-                              float* __restrict__ dw_f32, int rows, int C) {
+                              float* __restrict__ dw_f32, long rows, int C) {
   extern __shared__ float smem[];  // [C] dw accumulator + 16 scratch
   float* dw_local = smem;
   float* scratch = smem + C;
   for (int c = threadIdx.x; c < C; c += blockDim.x) dw_local[c] = 0.f;
   __syncthreads();
 
-  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
-    const T* dyr = dy + (long)row * C;
-    const T* xr = x + (long)row * C;
-    T* dxr = dx + (long)row * C;
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* dyr = dy + row * C;
+    const T* xr = x + row * C;
+    T* dxr = dx + row * C;
     const float mu = mean[row], rs = rstd[row];
     float s1 = 0.f, s2 = 0.f;
     for (int c = threadIdx.x; c < C; c += blockDim.x) {
@@ -95,17 +187,25 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, double eps) {
   auto mean = at::empty({rows}, x.options().dtype(at::kFloat));
   auto rstd = at::empty({rows}, x.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  const int block = 256;
-  const int grid = int(std::min<long>(rows, 2048));
-  if (x.scalar_type() == at::kBFloat16) {
-    hipLaunchKernelGGL(ln_fwd_kernel<uint16_t>, dim3(grid), dim3(block), 0, stream,
+  const bool vec = x.scalar_type() == at::kBFloat16 && C % 8 == 0 && C / 8 >= 64 &&
+                   C / 8 <= 1024;
+  if (vec) {
+    const int block = C / 8;
+    const int grid = int(std::min<long>(rows, 4096));
+    hipLaunchKernelGGL(ln_fwd_vec, dim3(grid), dim3(block), 0, stream,
                        (const uint16_t*)x.data_ptr(), (const uint16_t*)w.data_ptr(),
                        (uint16_t*)y.data_ptr(), mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), rows, C, (float)eps);
-  } else if (x.scalar_type() == at::kFloat) {
-    hipLaunchKernelGGL(ln_fwd_kernel<float>, dim3(grid), dim3(block), 0, stream,
-                       x.data_ptr<float>(), w.data_ptr<float>(), y.data_ptr<float>(),
+  } else if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(ln_fwd_kernel<uint16_t>, dim3(int(std::min<long>(rows, 2048))),
+                       dim3(256), 0, stream, (const uint16_t*)x.data_ptr(),
+                       (const uint16_t*)w.data_ptr(), (uint16_t*)y.data_ptr(),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(), rows, C, (float)eps);
+  } else if (x.scalar_type() == at::kFloat) {
+    hipLaunchKernelGGL(ln_fwd_kernel<float>, dim3(int(std::min<long>(rows, 2048))),
+                       dim3(256), 0, stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                       y.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       rows, C, (float)eps);
   } else {
     TORCH_CHECK(false, "layernorm: unsupported dtype");
   }
@@ -119,20 +219,32 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   auto dx = at::empty_like(x);
   auto dw_f32 = at::zeros({C}, x.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  const int block = 256;
-  const int grid = int(std::min<long>(rows, 1024));
-  const size_t smem = (C + 16) * sizeof(float);
-  if (x.scalar_type() == at::kBFloat16) {
-    hipLaunchKernelGGL(ln_bwd_kernel<uint16_t>, dim3(grid), dim3(block), smem, stream,
+  const bool vec = x.scalar_type() == at::kBFloat16 && C % 8 == 0 && C / 8 >= 64 &&
+                   C / 8 <= 1024;
+  if (vec) {
+    const int block = C / 8;
+    const int grid = int(std::min<long>(rows, 2048));
+    hipLaunchKernelGGL(ln_bwd_vec, dim3(grid), dim3(block), 0, stream,
                        (const uint16_t*)dy.data_ptr(), (const uint16_t*)x.data_ptr(),
                        (const uint16_t*)w.data_ptr(), mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), (uint16_t*)dx.data_ptr(),
                        dw_f32.data_ptr<float>(), rows, C);
   } else {
-    hipLaunchKernelGGL(ln_bwd_kernel<float>, dim3(grid), dim3(block), smem, stream,
-                       dy.data_ptr<float>(), x.data_ptr<float>(), w.data_ptr<float>(),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr<float>(),
-                       dw_f32.data_ptr<float>(), rows, C);
+    const int block = 256;
+    const int grid = int(std::min<long>(rows, 1024));
+    const size_t smem = (C + 16) * sizeof(float);
+    if (x.scalar_type() == at::kBFloat16) {
+      hipLaunchKernelGGL(ln_bwd_kernel<uint16_t>, dim3(grid), dim3(block), smem, stream,
+                         (const uint16_t*)dy.data_ptr(), (const uint16_t*)x.data_ptr(),
+                         (const uint16_t*)w.data_ptr(), mean.data_ptr<float>(),
+                         rstd.data_ptr<float>(), (uint16_t*)dx.data_ptr(),
+                         dw_f32.data_ptr<float>(), rows, C);
+    } else {
+      hipLaunchKernelGGL(ln_bwd_kernel<float>, dim3(grid), dim3(block), smem, stream,
+                         dy.data_ptr<float>(), x.data_ptr<float>(), w.data_ptr<float>(),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr<float>(),
+                         dw_f32.data_ptr<float>(), rows, C);
+    }
   }
   return {dx, dw_f32.to(x.scalar_type())};
 }
