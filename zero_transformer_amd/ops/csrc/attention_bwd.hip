@@ -182,11 +182,9 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
       // dQ += dS K : B[k=key][j=d] via tr-read of the row-major K tile
 #pragma unroll
       for (int d = 0; d < DB; ++d) {
-#pragma unroll
-        for (int s2 = 0; s2 < 2; ++s2) {
-          bf16x8 kb = tr_frag(k_lds, sub * 32 + s2 * 16, d * 32);
-          dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[s2], kb, dq_acc[d], 0, 0, 0);
-        }
+        attn::TrPair kp = attn::tr_frag_pair(k_lds, sub * 32, d * 32);
+        dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[0], kp.a, dq_acc[d], 0, 0, 0);
+        dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[1], kp.b, dq_acc[d], 0, 0, 0);
       }
     }
   }
@@ -338,13 +336,12 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
       // dV += P^T dO ; dK += dS^T Q : B[k=q][j=d] via tr-reads
 #pragma unroll
       for (int d = 0; d < DB; ++d) {
-#pragma unroll
-        for (int s2 = 0; s2 < 2; ++s2) {
-          bf16x8 dob = tr_frag(do_lds, sub * 32 + s2 * 16, d * 32);
-          bf16x8 qb = tr_frag(q_lds, sub * 32 + s2 * 16, d * 32);
-          dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[s2], dob, dv_acc[d], 0, 0, 0);
-          dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[s2], qb, dk_acc[d], 0, 0, 0);
-        }
+        attn::TrPair dop = attn::tr_frag_pair(do_lds, sub * 32, d * 32);
+        dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], dop.a, dv_acc[d], 0, 0, 0);
+        dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], dop.b, dv_acc[d], 0, 0, 0);
+        attn::TrPair qp = attn::tr_frag_pair(q_lds, sub * 32, d * 32);
+        dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[0], qp.a, dk_acc[d], 0, 0, 0);
+        dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[1], qp.b, dk_acc[d], 0, 0, 0);
       }
     }
   }

@@ -184,21 +184,9 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
       // ---- O += P @ V : V^T B-fragments via tr-read of row-major V ----
 #pragma unroll
       for (int d = 0; d < DB; ++d) {
-#pragma unroll
-        for (int s2 = 0; s2 < 2; ++s2) {
-#ifdef ZTA_FWD_NO_TR
-          bf16x8 b_frag;
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int row = sub * 32 + s2 * 16 + 8 * hi + e;
-            b_frag[e] = *(const __bf16*)((char*)v_lds +
-                                         swz(row, row * 256 + (d * 32 + li) * 2));
-          }
-#else
-          bf16x8 b_frag = tr_frag(v_lds, sub * 32 + s2 * 16, d * 32);
-#endif
-          o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[s2], b_frag, o_acc[d], 0, 0, 0);
-        }
+        attn::TrPair vp = attn::tr_frag_pair(v_lds, sub * 32, d * 32);
+        o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], vp.a, o_acc[d], 0, 0, 0);
+        o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], vp.b, o_acc[d], 0, 0, 0);
       }
     }
   }

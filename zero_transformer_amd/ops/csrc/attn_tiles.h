@@ -29,9 +29,11 @@ ZTA_DEV int swz(int row, int byte_off) { return byte_off ^ ((row & 7) << 4); }
 // see a ds_read inside inline asm, so it inserts no lgkmcnt wait before uses
 // of its outputs, and a separate waitcnt asm has no dataflow edge to the
 // outputs — the scheduler may move the consuming MFMA between read and wait
-// (observed: register-junk O values with exact lse). Bundling makes every
-// consumer order after the wait by SSA construction; "=&v" keeps the
-// destinations from aliasing the still-live address operands.
+// (observed: register-junk O values with exact lse). The "memory" clobber is
+// equally required: without it the compiler neither orders other waves'
+// staging stores against the asm read nor keeps them alive (no visible
+// reader). "=&v" keeps destinations from aliasing still-live address
+// operands.
 ZTA_DEV bf16x8 tr_frag(const uint16_t* lds, int k0, int j0) {
   const int l = threadIdx.x & 63;
   const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
@@ -49,10 +51,40 @@ ZTA_DEV bf16x8 tr_frag(const uint16_t* lds, int k0, int j0) {
       "s_waitcnt lgkmcnt(0)"
       : "=&v"(u.d[0]), "=&v"(u.d[1])
       : "v"(a0), "v"(a1)
-      : "memory");  // the asm READS LDS: without the clobber the compiler
-                    // neither orders other waves' staging stores against it
-                    // nor keeps them alive (no visible reader)
+      : "memory");
   return u.v;
+}
+
+// Two B-fragments (k0..k0+15 and k0+16..k0+31) in one asm statement: four
+// transpose reads and a single lgkmcnt drain, halving the per-fragment wait
+// + scheduling-barrier cost of tr_frag in the MFMA accumulation loops.
+struct TrPair {
+  bf16x8 a, b;
+};
+ZTA_DEV TrPair tr_frag_pair(const uint16_t* lds, int k0, int j0) {
+  const int l = threadIdx.x & 63;
+  const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
+  const int rb = 8 * (l >> 5) + ((l >> 2) & 3);
+  int a[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int row = k0 + 16 * (i >> 1) + 4 * (i & 1) + rb;
+    a[i] = (int)(size_t)((const char*)lds + row * 256 + (colb ^ ((row & 7) << 4)));
+  }
+  union {
+    i32x2 d[4];
+    TrPair f;
+  } u;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n\t"
+      "ds_read_b64_tr_b16 %1, %5\n\t"
+      "ds_read_b64_tr_b16 %2, %6\n\t"
+      "ds_read_b64_tr_b16 %3, %7\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(u.d[0]), "=&v"(u.d[1]), "=&v"(u.d[2]), "=&v"(u.d[3])
+      : "v"(a[0]), "v"(a[1]), "v"(a[2]), "v"(a[3])
+      : "memory");
+  return u.f;
 }
 
 // In-register C-layout -> A-fragment transform (T12): 16 f32 values x[r]

@@ -2,8 +2,10 @@
 // (N, 50304) one-hot of the reference (GPT.py:105-111, losses.py:10-23).
 // fp32 log-softmax statistics over bf16 logits.
 //
-// fwd: one block per row -> per-row loss (lse - logit[target]) + saved lse.
-// bwd: dlogits = (softmax - onehot) * dloss / N  in one vectorized pass.
+// fwd: one block per row, single pass over the logits (online softmax:
+//      running max + rescaled sum), s16x8-vectorized loads (G13) ->
+//      per-row loss (lse - logit[target]) + saved lse.
+// bwd: dlogits = (softmax - onehot) * dloss / N in one vectorized pass.
 
 #include "common.h"
 
@@ -14,6 +16,86 @@
 
 namespace {
 
+// bf16 fast path: V % 8 == 0. Online max/sum in a single read of the row.
+__global__ __launch_bounds__(256) void ce_fwd_vec(
+    const uint16_t* __restrict__ logits, const long* __restrict__ targets,
+    float* __restrict__ loss, float* __restrict__ lse_out, long rows, int V) {
+  __shared__ float scratch[16];
+  const int t = threadIdx.x;
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const uint16_t* lr = logits + row * V;
+    float m = -3.4e38f, s = 0.f;
+    for (int c = t * 8; c < V; c += 256 * 8) {
+      s16x8 x8 = *reinterpret_cast<const s16x8*>(&lr[c]);
+      float xv[8], lm = -3.4e38f;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        xv[e] = bf16_to_f32((uint16_t)x8[e]);
+        lm = fmaxf(lm, xv[e]);
+      }
+      if (lm > m) {
+        s *= __expf(m - lm);
+        m = lm;
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e) s += __expf(xv[e] - m);
+    }
+    // cross-lane/wave combine of (m, s)
+    {
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) {
+        float mo = __shfl_xor(m, off, WAVE);
+        float so = __shfl_xor(s, off, WAVE);
+        float mn = fmaxf(m, mo);
+        s = s * __expf(m - mn) + so * __expf(mo - mn);
+        m = mn;
+      }
+      const int lane = t & (WAVE - 1), wid = t / WAVE;
+      __shared__ float sm[4], ss[4];
+      if (lane == 0) {
+        sm[wid] = m;
+        ss[wid] = s;
+      }
+      __syncthreads();
+      float mn = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+      s = ss[0] * __expf(sm[0] - mn) + ss[1] * __expf(sm[1] - mn) +
+          ss[2] * __expf(sm[2] - mn) + ss[3] * __expf(sm[3] - mn);
+      m = mn;
+      __syncthreads();
+    }
+    const float lse = m + __logf(s);
+    if (t == 0) {
+      lse_out[row] = lse;
+      loss[row] = lse - bf16_to_f32(lr[targets[row]]);
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void ce_bwd_vec(
+    const uint16_t* __restrict__ logits, const long* __restrict__ targets,
+    const float* __restrict__ lse, const float* __restrict__ dloss,
+    uint16_t* __restrict__ dlogits, long rows, int V) {
+  const float scale = dloss[0] / rows;
+  const int t = threadIdx.x;
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const uint16_t* lr = logits + row * V;
+    uint16_t* dr = dlogits + row * V;
+    const float l = lse[row];
+    const long tg = targets[row];
+    for (int c = t * 8; c < V; c += 256 * 8) {
+      s16x8 x8 = *reinterpret_cast<const s16x8*>(&lr[c]);
+      s16x8 d8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float p = __expf(bf16_to_f32((uint16_t)x8[e]) - l);
+        d8[e] = (short)f32_to_bf16(scale * (p - (c + e == tg ? 1.f : 0.f)));
+      }
+      *reinterpret_cast<s16x8*>(&dr[c]) = d8;
+    }
+  }
+}
+
+// Generic fallback (fp32 or odd V)
 template <typename T>
 __global__ void ce_fwd_kernel(const T* __restrict__ logits,
                               const long* __restrict__ targets,
@@ -79,8 +161,12 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets)
   auto lse = at::empty({rows}, logits.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int block = 256;
-  const int grid = int(std::min<long>(rows, 2048));
-  if (logits.scalar_type() == at::kBFloat16) {
+  const int grid = int(std::min<long>(rows, 4096));
+  if (logits.scalar_type() == at::kBFloat16 && V % 8 == 0) {
+    hipLaunchKernelGGL(ce_fwd_vec, dim3(grid), dim3(block), 0, stream,
+                       (const uint16_t*)logits.data_ptr(), targets.data_ptr<long>(),
+                       loss.data_ptr<float>(), lse.data_ptr<float>(), rows, V);
+  } else if (logits.scalar_type() == at::kBFloat16) {
     hipLaunchKernelGGL(ce_fwd_kernel<uint16_t>, dim3(grid), dim3(block), 0, stream,
                        (const uint16_t*)logits.data_ptr(), targets.data_ptr<long>(),
                        loss.data_ptr<float>(), lse.data_ptr<float>(), rows, V);
@@ -102,8 +188,13 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor l
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   auto dl = dloss.to(at::kFloat).contiguous();
   const int block = 256;
-  const int grid = int(std::min<long>(rows, 2048));
-  if (logits.scalar_type() == at::kBFloat16) {
+  const int grid = int(std::min<long>(rows, 4096));
+  if (logits.scalar_type() == at::kBFloat16 && V % 8 == 0) {
+    hipLaunchKernelGGL(ce_bwd_vec, dim3(grid), dim3(block), 0, stream,
+                       (const uint16_t*)logits.data_ptr(), targets.data_ptr<long>(),
+                       lse.data_ptr<float>(), dl.data_ptr<float>(),
+                       (uint16_t*)dlogits.data_ptr(), rows, V);
+  } else if (logits.scalar_type() == at::kBFloat16) {
     hipLaunchKernelGGL(ce_bwd_kernel<uint16_t>, dim3(grid), dim3(block), 0, stream,
                        (const uint16_t*)logits.data_ptr(), targets.data_ptr<long>(),
                        lse.data_ptr<float>(), dl.data_ptr<float>(),
