@@ -93,7 +93,7 @@ def test_weight_tying():
 def test_init_scaling():
     model = GPT(DotDict(dict(CFG, N=8)))
     resid = model.blocks[0].attn.fc_resid.weight
-    base = model.blocks[0].attn.query.weight
+    base = model.blocks[0].attn.qkv_w
     # residual-out init std is 0.02/sqrt(2N)
     assert resid.std().item() < base.std().item()
     assert abs(resid.std().item() - 0.02 / math.sqrt(16)) < 2e-3

@@ -43,18 +43,24 @@ class LayerNorm(nn.Module):
 
 
 class CausalSelfAttention(nn.Module):
-    """Causal multi-head attention with optional ALiBi position biasing."""
+    """Causal multi-head attention with optional ALiBi position biasing.
+
+    The q/k/v projections run as ONE fused GEMM (qkv_w, (3C, C)) — one pass
+    over x and a bigger, better-shaped hipBLASLt GEMM than three (C, C)
+    calls. The .pth contract (separate query/key/value weights,
+    flax_to_pytorch.py:10-35) is preserved at every state_dict boundary by
+    the _save/_load hooks below.
+    """
 
     def __init__(self, cfg):
         super().__init__()
         dim, heads = cfg.embedding_dim, cfg.num_head
         assert dim % heads == 0
+        self.dim = dim
         self.num_head = heads
         self.head_dim = dim // heads
         self.dropout_p = cfg.dropout
-        self.query = nn.Linear(dim, dim, bias=False)
-        self.key = nn.Linear(dim, dim, bias=False)
-        self.value = nn.Linear(dim, dim, bias=False)
+        self.qkv_w = nn.Parameter(torch.empty(3 * dim, dim))
         self.fc_resid = nn.Linear(dim, dim, bias=False)
         if cfg.alibi_attn:
             self.register_buffer("slopes", ops.alibi_slopes(heads), persistent=False)
@@ -65,14 +71,33 @@ class CausalSelfAttention(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, T, C = x.shape
         H, D = self.num_head, self.head_dim
-        q = self.query(x).view(B, T, H, D).transpose(1, 2)
-        k = self.key(x).view(B, T, H, D).transpose(1, 2)
-        v = self.value(x).view(B, T, H, D).transpose(1, 2)
+        qkv = F.linear(x, self.qkv_w)
+        q, k, v = (
+            t.view(B, T, H, D).transpose(1, 2) for t in qkv.split(C, dim=-1)
+        )
         o = ops.attention(
             q, k, v, self.slopes, dropout_p=self.dropout_p, training=self.training
         )
         o = o.transpose(1, 2).reshape(B, T, C)
         return self.resid_dropout(self.fc_resid(o))
+
+    # .pth contract: expose query/key/value instead of the fused qkv_w
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+        w = destination.pop(prefix + "qkv_w")
+        C = self.dim
+        destination[prefix + "query.weight"] = w[:C]
+        destination[prefix + "key.weight"] = w[C : 2 * C]
+        destination[prefix + "value.weight"] = w[2 * C :]
+
+    def _load_from_state_dict(self, state_dict, prefix, *args, **kwargs):
+        C = self.dim
+        names = [prefix + n + ".weight" for n in ("query", "key", "value")]
+        if all(n in state_dict for n in names):
+            state_dict[prefix + "qkv_w"] = torch.cat(
+                [state_dict.pop(n) for n in names], dim=0
+            )
+        super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
 
 
 class MLP(nn.Module):

@@ -75,10 +75,10 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
   uint16_t* k_lds = (uint16_t*)smem;
   uint16_t* v_lds = k_lds + TB * 128;
 
-  const int bh = blockIdx.y;
+  const int bh = blockIdx.x;  // grid: (BH, tiles) for per-CU load balance
   const int h = bh % H;
   const long base = (long)bh * T * D;
-  const int q0 = blockIdx.x * RB;
+  const int q0 = blockIdx.y * RB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int li = lane & 31, hi = lane >> 5;
@@ -219,10 +219,10 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
   float* lse_s = (float*)(do_lds + TB * 128);
   float* delta_s = lse_s + TB;
 
-  const int bh = blockIdx.y;
+  const int bh = blockIdx.x;  // grid: (BH, tiles) for per-CU load balance
   const int h = bh % H;
   const long base = (long)bh * T * D;
-  const int k0 = blockIdx.x * RB;
+  const int k0 = blockIdx.y * RB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int li = lane & 31, hi = lane >> 5;
@@ -366,7 +366,7 @@ void launch_bwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
                 const at::Tensor& slopes, at::Tensor& dq, at::Tensor& dk,
                 at::Tensor& dv, int B, int H, int T, float scale, float p_drop,
                 uint32_t seed, hipStream_t stream) {
-  dim3 grid((T + RB - 1) / RB, B * H);
+  dim3 grid(B * H, (T + RB - 1) / RB);
   const size_t smem_dq = 2 * TB * 128 * sizeof(uint16_t);
   hipLaunchKernelGGL(flash_dq_kernel<D>, grid, dim3(512), smem_dq, stream,
                      (const uint16_t*)q.data_ptr(), (const uint16_t*)k.data_ptr(),

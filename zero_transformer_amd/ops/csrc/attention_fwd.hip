@@ -54,10 +54,11 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
   uint16_t* k_lds = (uint16_t*)smem;       // 64 x [256 B], swizzled
   uint16_t* v_lds = k_lds + TB * 128;      // 64 x [256 B], swizzled
 
-  const int bh = blockIdx.y;
+  const int bh = blockIdx.x;  // grid: (BH, tiles) — consecutive blocks share the
+  // tile index so per-CU work is balanced across the causal triangle
   const int h = bh % H;
   const long base = (long)bh * T * D;
-  const int q0 = blockIdx.x * RB;
+  const int q0 = blockIdx.y * RB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int li = lane & 31;
@@ -212,7 +213,7 @@ void launch_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
                 const at::Tensor& slopes, at::Tensor& o, at::Tensor& lse, int B,
                 int H, int T, float scale, float p_drop, uint32_t seed,
                 hipStream_t stream) {
-  dim3 grid((T + RB - 1) / RB, B * H);
+  dim3 grid(B * H, (T + RB - 1) / RB);
   const size_t smem = 2 * TB * 128 * sizeof(uint16_t);
   hipLaunchKernelGGL(flash_fwd_kernel<D>, grid, dim3(512), smem, stream,
                      (const uint16_t*)q.data_ptr(), (const uint16_t*)k.data_ptr(),

@@ -231,13 +231,27 @@ class ZeRO1Optimizer:
     # ------------------------------------------------------------------
     @torch.no_grad()
     def full_param_state_dict(self) -> Dict[str, torch.Tensor]:
-        """Gather fp32 master params to a full state_dict (rank 0 usable)."""
+        """Gather fp32 master params to a full state_dict (rank 0 usable).
+
+        Name convention: a parameter `<m>.qkv_w` (the fused q/k/v projection
+        of models.gpt.CausalSelfAttention) is exported as the three separate
+        `<m>.{query,key,value}.weight` tensors so the result follows the
+        torch_compatability .pth contract (flax_to_pytorch.py:10-35).
+        """
         out: Dict[str, torch.Tensor] = {}
         for b in self.buckets:
             full = torch.zeros(b.numel, dtype=torch.float32, device=self.device)
             comm.all_gather_flat(full, b.master)
             for n, p, o in zip(b.names, b.params, b.offsets):
-                out[n] = full[o : o + p.numel()].view(p.shape).cpu().clone()
+                t = full[o : o + p.numel()].view(p.shape).cpu().clone()
+                if n.endswith("qkv_w"):
+                    base = n[: -len("qkv_w")]
+                    C = t.shape[0] // 3
+                    out[base + "query.weight"] = t[:C]
+                    out[base + "key.weight"] = t[C : 2 * C]
+                    out[base + "value.weight"] = t[2 * C :]
+                else:
+                    out[n] = t
             del full
         return out
 
@@ -258,12 +272,23 @@ class ZeRO1Optimizer:
 
     @torch.no_grad()
     def load_param_state_dict(self, sd: Dict[str, torch.Tensor]):
-        """Load full fp32 params: refill master shards + bf16 working copy."""
+        """Load full fp32 params: refill master shards + bf16 working copy.
+
+        Accepts the .pth layout (separate query/key/value) for fused qkv_w
+        params — see full_param_state_dict.
+        """
         shard = lambda t, n: t[self.rank * (n // self.world) : (self.rank + 1) * (n // self.world)]
         for b in self.buckets:
             full = torch.zeros(b.numel, dtype=torch.float32, device=self.device)
             for n, p, o in zip(b.names, b.params, b.offsets):
-                full[o : o + p.numel()].copy_(sd[n].reshape(-1).float().to(self.device))
+                if n.endswith("qkv_w") and n not in sd:
+                    base = n[: -len("qkv_w")]
+                    t = torch.cat(
+                        [sd[base + k + ".weight"] for k in ("query", "key", "value")], dim=0
+                    )
+                else:
+                    t = sd[n]
+                full[o : o + p.numel()].copy_(t.reshape(-1).float().to(self.device))
             b.master.copy_(shard(full, b.numel))
             b.flat_param.copy_(full.to(self.param_dtype))
             del full
