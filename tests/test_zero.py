@@ -26,7 +26,10 @@ def test_param_views_share_storage():
     for b in opt.buckets:
         for p, o in zip(b.params, b.offsets):
             assert p.data.data_ptr() == b.flat_param[o:].data_ptr()
-            assert p.grad.data_ptr() == b.flat_grad[o:].data_ptr()
+            # grads start as None (assigned by autograd, adopted into the
+            # flat bucket view by the post-accumulate hook)
+            assert p.grad is None
+            assert opt._grad_view[id(p)].data_ptr() == b.flat_grad[o:].data_ptr()
 
 
 def test_decay_grouping():

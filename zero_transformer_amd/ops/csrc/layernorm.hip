@@ -60,13 +60,16 @@ __global__ __launch_bounds__(1024) void ln_fwd_vec(
 
 // dx = rstd * (g - mean(g) - xhat * mean(g*xhat)),  g = w * dy
 // dw[c] = sum_rows dy[r,c] * xhat[r,c]  — thread-owned columns, register
-// accumulation across the row loop, one atomicAdd per column at the end.
+// accumulation across the row loop; per-block partial rows written to
+// dw_part[block][C] (no atomics — an atomic tail at grid 2048 was 4.2M
+// atomicAdds onto 2048 words and dominated the kernel), reduced by
+// ln_dw_reduce. The two row statistics reduce together (one barrier set).
 __global__ __launch_bounds__(1024) void ln_bwd_vec(
     const uint16_t* __restrict__ dy, const uint16_t* __restrict__ x,
     const uint16_t* __restrict__ w, const float* __restrict__ mean,
     const float* __restrict__ rstd, uint16_t* __restrict__ dx,
-    float* __restrict__ dw_f32, long rows, int C) {
-  __shared__ float scratch[16];
+    float* __restrict__ dw_part, long rows, int C) {
+  __shared__ float scratch[32];
   const int t = threadIdx.x;
   float wv[8], dw_acc[8];
   {
@@ -78,6 +81,7 @@ __global__ __launch_bounds__(1024) void ln_bwd_vec(
     }
   }
   const float invC = 1.0f / (float)C;
+  const int lane = t & (WAVE - 1), wid = t / WAVE, nw = blockDim.x / WAVE;
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     s16x8 dy8 = *reinterpret_cast<const s16x8*>(&dy[row * C + t * 8]);
     s16x8 x8 = *reinterpret_cast<const s16x8*>(&x[row * C + t * 8]);
@@ -93,16 +97,45 @@ __global__ __launch_bounds__(1024) void ln_bwd_vec(
       s2 += g[e] * xh[e];
       dw_acc[e] += d * xh[e];
     }
-    s1 = block_reduce_sum(s1, scratch) * invC;
-    s2 = block_reduce_sum(s2, scratch) * invC;
+    // joint block reduce of (s1, s2): one LDS round instead of two
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      s1 += __shfl_xor(s1, off, WAVE);
+      s2 += __shfl_xor(s2, off, WAVE);
+    }
+    if (lane == 0) {
+      scratch[wid] = s1;
+      scratch[16 + wid] = s2;
+    }
+    __syncthreads();
+    s1 = 0.f;
+    s2 = 0.f;
+    for (int i = 0; i < nw; ++i) {
+      s1 += scratch[i];
+      s2 += scratch[16 + i];
+    }
+    __syncthreads();
+    s1 *= invC;
+    s2 *= invC;
     s16x8 dx8;
 #pragma unroll
     for (int e = 0; e < 8; ++e)
       dx8[e] = (short)f32_to_bf16((g[e] - s1 - xh[e] * s2) * rs);
     *reinterpret_cast<s16x8*>(&dx[row * C + t * 8]) = dx8;
   }
-#pragma unroll
-  for (int e = 0; e < 8; ++e) atomicAdd(&dw_f32[t * 8 + e], dw_acc[e]);
+  f32x4 p0 = {dw_acc[0], dw_acc[1], dw_acc[2], dw_acc[3]};
+  f32x4 p1 = {dw_acc[4], dw_acc[5], dw_acc[6], dw_acc[7]};
+  *reinterpret_cast<f32x4*>(&dw_part[(long)blockIdx.x * C + t * 8]) = p0;
+  *reinterpret_cast<f32x4*>(&dw_part[(long)blockIdx.x * C + t * 8 + 4]) = p1;
+}
+
+__global__ void ln_dw_reduce(const float* __restrict__ dw_part,
+                             float* __restrict__ dw, int C, int npart) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f;
+  for (int b = 0; b < npart; ++b) s += dw_part[(long)b * C + c];
+  dw[c] = s;
 }
 
 // ---------------------------------------------------------------------------
@@ -223,12 +256,15 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
                    C / 8 <= 1024;
   if (vec) {
     const int block = C / 8;
-    const int grid = int(std::min<long>(rows, 2048));
+    const int grid = int(std::min<long>(rows, 512));
+    auto dw_part = at::empty({grid, C}, x.options().dtype(at::kFloat));
     hipLaunchKernelGGL(ln_bwd_vec, dim3(grid), dim3(block), 0, stream,
                        (const uint16_t*)dy.data_ptr(), (const uint16_t*)x.data_ptr(),
                        (const uint16_t*)w.data_ptr(), mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), (uint16_t*)dx.data_ptr(),
-                       dw_f32.data_ptr<float>(), rows, C);
+                       dw_part.data_ptr<float>(), rows, C);
+    hipLaunchKernelGGL(ln_dw_reduce, dim3((C + 255) / 256), dim3(256), 0, stream,
+                       dw_part.data_ptr<float>(), dw_f32.data_ptr<float>(), C, grid);
   } else {
     const int block = 256;
     const int grid = int(std::min<long>(rows, 1024));

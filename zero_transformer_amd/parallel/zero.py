@@ -154,11 +154,12 @@ class ZeRO1Optimizer:
         return b
 
     def _install_views_and_hooks(self):
+        self._grad_view: Dict[int, torch.Tensor] = {}
         for b in self.buckets:
             for p, o in zip(b.params, b.offsets):
                 shape = p.data.shape
                 p.data = b.flat_param[o : o + p.numel()].view(shape)
-                p.grad = b.flat_grad[o : o + p.numel()].view(shape)
+                self._grad_view[id(p)] = b.flat_grad[o : o + p.numel()].view(shape)
                 p.register_post_accumulate_grad_hook(self._on_grad_ready)
 
     # ------------------------------------------------------------------
@@ -167,6 +168,15 @@ class ZeRO1Optimizer:
         self._sync = sync
 
     def _on_grad_ready(self, p: torch.nn.Parameter):
+        # First accumulation of the step: adopt autograd's freshly-assigned
+        # grad by copying it into the flat bucket view and re-pointing
+        # p.grad at the view, so later micro-steps accumulate in place.
+        # This replaces the old preset-view scheme, which paid a bucket-wide
+        # zero_() plus a read-modify-write add on the FIRST backward too.
+        view = self._grad_view[id(p)]
+        if p.grad is not view:
+            view.copy_(p.grad)
+            p.grad = view
         if not self._sync:
             return
         b, _ = self._param_bucket[id(p)]
@@ -188,9 +198,15 @@ class ZeRO1Optimizer:
         assert self._sync, "step() called while grad sync disabled"
         self.step_count += 1
         lr = float(self.lr(self.step_count))
-        # any bucket whose hook never fired (e.g. unused param) — reduce now
+        # any bucket whose hook never fired (e.g. unused param) — zero the
+        # stale regions (the flat buffer holds last step's values until a
+        # hook adopts a fresh grad) and reduce now
         for b in self.buckets:
             if b.ready != len(b.params):
+                for p in b.params:
+                    view = self._grad_view[id(p)]
+                    if p.grad is not view:
+                        view.zero_()
                 self._launch_reduce(b)
         if self.overlap_comm:
             torch.cuda.current_stream(self.device).wait_stream(self._comm_stream)
@@ -221,8 +237,11 @@ class ZeRO1Optimizer:
 
     @torch.no_grad()
     def zero_grad(self, set_to_none: bool = False):
+        # p.grad = None: the next backward ASSIGNS fresh grads (no add into
+        # zeroed memory); _on_grad_ready copies them into the flat buckets.
         for b in self.buckets:
-            b.flat_grad.zero_()
+            for p in b.params:
+                p.grad = None
             b.ready = 0
             b.rs_work = None
 
