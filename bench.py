@@ -48,7 +48,10 @@ def main():
     from zero_transformer_amd.models import model_getter
     from zero_transformer_amd.parallel.zero import ZeRO1Optimizer
     from zero_transformer_amd.training.trainer import TrainEngine
+    from zero_transformer_amd.utils import gemm_tune
     from zero_transformer_amd.utils.lr import warmup_cosine
+
+    gemm_tune.enable()  # committed hipBLASLt tunings (no-op if absent)
 
     torch.manual_seed(1234 + rank)
     model, mcfg = model_getter(args.model, return_cfg=True)

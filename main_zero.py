@@ -68,6 +68,10 @@ def main():
     cfg = load_config(args.cfg)
     device, rank, world = init_distributed()
     t_cfg = cfg.training
+    if device.type == "cuda":
+        from zero_transformer_amd.utils import gemm_tune
+
+        gemm_tune.enable()  # committed hipBLASLt tunings (no-op if absent)
 
     model, model_cfg = model_getter(
         cfg.model.size, config_path=args.model_cfg, return_cfg=True

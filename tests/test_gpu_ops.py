@@ -178,10 +178,11 @@ def test_attention_bwd_with_dropout_grad_matches_fd(dev):
     q, k, v = (torch.randn(B, H, T, D, device=dev).to(torch.bfloat16) for _ in range(3))
     slopes = reference.alibi_slopes(H).to(dev)
     ext = O.hip_ops()
-    o1, lse1 = ext.attn_fwd(q, k, v, slopes, 0.3, 42)
-    o2, lse2 = ext.attn_fwd(q, k, v, slopes, 0.3, 42)
+    qkv = torch.cat([t.transpose(1, 2).reshape(B, T, H * D) for t in (q, k, v)], -1).contiguous()
+    o1, lse1 = ext.attn_fwd(qkv, slopes, H, 0.3, 42)
+    o2, lse2 = ext.attn_fwd(qkv, slopes, H, 0.3, 42)
     assert torch.equal(o1, o2), "same seed must give identical dropout output"
-    o3, _ = ext.attn_fwd(q, k, v, slopes, 0.3, 43)
+    o3, _ = ext.attn_fwd(qkv, slopes, H, 0.3, 43)
     assert not torch.equal(o1, o3), "different seed must change dropout output"
 
 
@@ -200,9 +201,14 @@ def test_attention_dropout_fwd_bwd_exact_mask_parity(dev, p):
     q, k, v = (torch.randn(B, H, T, D, device=dev).to(torch.bfloat16) for _ in range(3))
     slopes = reference.alibi_slopes(H).to(dev)
     ext = O.hip_ops()
-    o, lse = ext.attn_fwd(q, k, v, slopes, p, seed)
+    C = H * D
+    qkv = torch.cat([t.transpose(1, 2).reshape(B, T, C) for t in (q, k, v)], -1).contiguous()
+    o, lse = ext.attn_fwd(qkv, slopes, H, p, seed)
     do = torch.randn_like(o)
-    dq, dk, dv = ext.attn_bwd(do, q, k, v, slopes, o, lse, p, seed)
+    (dqkv,) = ext.attn_bwd(do, qkv, slopes, o, lse, H, p, seed)
+    dq, dk, dv = (t.view(B, T, H, D).transpose(1, 2) for t in dqkv.split(C, dim=-1))
+    o = o.view(B, T, H, D).transpose(1, 2)
+    do = do.view(B, T, H, D).transpose(1, 2)
 
     keep = reference.drop_mask(seed, B, H, T, p)
     qr, kr, vr = (t.detach().float().cpu().requires_grad_(True) for t in (q, k, v))

@@ -36,18 +36,18 @@ def main():
     B, H, T, D = args.B, args.H, args.T, args.D
     dev = torch.device("cuda")
     torch.manual_seed(0)
-    q, k, v = (torch.randn(B, H, T, D, device=dev).to(torch.bfloat16) for _ in range(3))
+    qkv = torch.randn(B, T, 3 * H * D, device=dev).to(torch.bfloat16)
     slopes = reference.alibi_slopes(H).to(dev)
     ext = ops.hip_ops()
 
     fwd_flops = 2 * 2 * B * H * (T * T / 2) * D  # QK^T + PV, causal half
-    t = bench(lambda: ext.attn_fwd(q, k, v, slopes, args.dropout, 7))
+    t = bench(lambda: ext.attn_fwd(qkv, slopes, H, args.dropout, 7))
     print(f"fwd : {t*1e3:8.3f} ms  {fwd_flops/t/1e12:7.1f} TF/s")
 
-    o, lse = ext.attn_fwd(q, k, v, slopes, args.dropout, 7)
+    o, lse = ext.attn_fwd(qkv, slopes, H, args.dropout, 7)
     do = torch.randn_like(o)
     bwd_flops = fwd_flops * 2.5  # 5 GEMMs vs 2
-    t = bench(lambda: ext.attn_bwd(do, q, k, v, slopes, o, lse, args.dropout, 7))
+    t = bench(lambda: ext.attn_bwd(do, qkv, slopes, o, lse, H, args.dropout, 7))
     print(f"bwd : {t*1e3:8.3f} ms  {bwd_flops/t/1e12:7.1f} TF/s")
 
 

@@ -17,9 +17,14 @@ def run(B, H, T, D):
     q, k, v = (torch.randn(B, H, T, D, device=dev).to(torch.bfloat16) for _ in range(3))
     slopes = reference.alibi_slopes(H).to(dev)
     ext = O.hip_ops()
-    o, lse = ext.attn_fwd(q, k, v, slopes, 0.0, 0)
+    C = H * D
+    qkv = torch.cat([t.transpose(1, 2).reshape(B, T, C) for t in (q, k, v)], -1).contiguous()
+    o, lse = ext.attn_fwd(qkv, slopes, H, 0.0, 0)
     do = torch.randn_like(o)
-    dq, dk, dv = ext.attn_bwd(do, q, k, v, slopes, o, lse, 0.0, 0)
+    (dqkv,) = ext.attn_bwd(do, qkv, slopes, o, lse, H, 0.0, 0)
+    dq, dk, dv = (t.view(B, T, H, D).transpose(1, 2).contiguous() for t in dqkv.split(C, dim=-1))
+    o = o.view(B, T, H, D).transpose(1, 2).contiguous()
+    do = do.view(B, T, H, D).transpose(1, 2)
 
     qr, kr, vr = (t.detach().float().requires_grad_(True) for t in (q, k, v))
     ref = reference.attention(qr, kr, vr, slopes)

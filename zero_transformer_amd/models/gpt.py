@@ -69,16 +69,11 @@ class CausalSelfAttention(nn.Module):
         self.resid_dropout = nn.Dropout(cfg.dropout)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        B, T, C = x.shape
-        H, D = self.num_head, self.head_dim
-        qkv = F.linear(x, self.qkv_w)
-        q, k, v = (
-            t.view(B, T, H, D).transpose(1, 2) for t in qkv.split(C, dim=-1)
+        qkv = F.linear(x, self.qkv_w)  # (B, T, 3C)
+        o = ops.attention_qkv(
+            qkv, self.num_head, self.slopes,
+            dropout_p=self.dropout_p, training=self.training,
         )
-        o = ops.attention(
-            q, k, v, self.slopes, dropout_p=self.dropout_p, training=self.training
-        )
-        o = o.transpose(1, 2).reshape(B, T, C)
         return self.resid_dropout(self.fc_resid(o))
 
     # .pth contract: expose query/key/value instead of the fused qkv_w

@@ -108,29 +108,62 @@ def gelu(x: torch.Tensor) -> torch.Tensor:
 
 
 class _FlashAttentionFn(torch.autograd.Function):
+    """Fused-layout flash attention: qkv (B, T, 3C) -> o (B, T, C).
+
+    The fused layout is the direct output of the model's single qkv GEMM;
+    backward emits dqkv in the same layout — no transposes, splits or cats
+    anywhere on the attention path.
+    """
+
     @staticmethod
-    def forward(ctx, q, k, v, slopes, dropout_p, training):
+    def forward(ctx, qkv, slopes, num_head, dropout_p, training):
         ext = hip_ops()
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        qkv = qkv.contiguous()
         if slopes is None:
-            slopes = torch.zeros(q.shape[1], dtype=torch.float32, device=q.device)
+            slopes = torch.zeros(num_head, dtype=torch.float32, device=qkv.device)
         if dropout_p > 0.0 and training:
             seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
         else:
             seed, dropout_p = 0, 0.0
-        o, lse = ext.attn_fwd(q, k, v, slopes, float(dropout_p), seed)
-        ctx.save_for_backward(q, k, v, slopes, o, lse)
+        o, lse = ext.attn_fwd(qkv, slopes, num_head, float(dropout_p), seed)
+        ctx.save_for_backward(qkv, slopes, o, lse)
+        ctx.num_head = num_head
         ctx.dropout_p = dropout_p
         ctx.seed = seed
         return o
 
     @staticmethod
     def backward(ctx, do):
-        q, k, v, slopes, o, lse = ctx.saved_tensors
-        dq, dk, dv = hip_ops().attn_bwd(
-            do.contiguous(), q, k, v, slopes, o, lse, ctx.dropout_p, ctx.seed
+        qkv, slopes, o, lse = ctx.saved_tensors
+        (dqkv,) = hip_ops().attn_bwd(
+            do.contiguous(), qkv, slopes, o, lse, ctx.num_head, ctx.dropout_p, ctx.seed
         )
-        return dq, dk, dv, None, None, None
+        return dqkv, None, None, None, None
+
+
+def attention_qkv(
+    qkv: torch.Tensor,
+    num_head: int,
+    slopes: Optional[torch.Tensor] = None,
+    dropout_p: float = 0.0,
+    training: bool = False,
+) -> torch.Tensor:
+    """Causal (ALiBi-biased) attention on the fused qkv projection.
+
+    qkv: (B, T, 3C) with head-interleaved columns (q | k | v, each (H, D)
+    blocks); returns (B, T, C). GPU path requires D % 32 == 0.
+    """
+    B, T, C3 = qkv.shape
+    C = C3 // 3
+    D = C // num_head
+    if qkv.is_cuda and D % 32 == 0:
+        return _FlashAttentionFn.apply(qkv, slopes, num_head, dropout_p, training)
+    # eager fallback (CPU / odd head_dim): unpack to (B, H, T, D)
+    q, k, v = (
+        t.view(B, T, num_head, D).transpose(1, 2) for t in qkv.split(C, dim=-1)
+    )
+    o = reference.attention(q, k, v, slopes, dropout_p, training)
+    return o.transpose(1, 2).reshape(B, T, C)
 
 
 def attention(
@@ -144,6 +177,8 @@ def attention(
 ) -> torch.Tensor:
     """Causal (ALiBi-biased) attention. q,k,v: (B, H, T, D) -> (B, H, T, D).
 
+    Convenience wrapper over attention_qkv (packs into the fused layout —
+    copies; the model's hot path calls attention_qkv directly).
     impl: "auto" (fused on GPU, eager on CPU), "fused", or "eager".
     """
     if impl == "auto":
@@ -158,7 +193,12 @@ def attention(
         )
         impl = "eager"
     if impl == "fused":
-        return _FlashAttentionFn.apply(q, k, v, slopes, dropout_p, training)
+        B, H, T, D = q.shape
+        qkv = torch.cat(
+            [t.transpose(1, 2).reshape(B, T, H * D) for t in (q, k, v)], dim=-1
+        )
+        o = attention_qkv(qkv, H, slopes, dropout_p, training)
+        return o.view(B, T, H, D).transpose(1, 2)
     return reference.attention(q, k, v, slopes, dropout_p, training)
 
 

@@ -44,16 +44,23 @@ using attn::tr_frag;
 constexpr float NEG_INF = -3.0e38f;
 
 // ---------------------------------------------------------------------------
+// dout/o are (B, T, C); delta is (B, H, T): row index r = (b*H + h)*T + t
+// maps to plane offset (b*T + t)*C + h*D.
 __global__ void delta_kernel(const uint16_t* __restrict__ dout,
                              const uint16_t* __restrict__ o,
-                             float* __restrict__ delta, long rows, int D) {
+                             float* __restrict__ delta, long rows, int H, int T,
+                             int C, int D) {
   const int wid = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int nw = gridDim.x * blockDim.x / WAVE;
   for (long row = wid; row < rows; row += nw) {
+    const long b = row / ((long)H * T);
+    const int h = (int)((row / T) % H);
+    const int t = (int)(row % T);
+    const long off = (b * T + t) * C + h * D;
     float s = 0.f;
     for (int d = lane; d < D; d += WAVE)
-      s += bf16_to_f32(dout[row * D + d]) * bf16_to_f32(o[row * D + d]);
+      s += bf16_to_f32(dout[off + d]) * bf16_to_f32(o[off + d]);
     s = wave_reduce_sum(s);
     if (lane == 0) delta[row] = s;
   }
@@ -66,18 +73,21 @@ __global__ void delta_kernel(const uint16_t* __restrict__ dout,
 // ---------------------------------------------------------------------------
 template <int D>
 __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
-    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
-    const uint16_t* __restrict__ v, const uint16_t* __restrict__ dout,
+    const uint16_t* __restrict__ qkv, const uint16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    const float* __restrict__ slopes, uint16_t* __restrict__ dq, int H, int T,
-    float scale, float p_drop, uint32_t seed) {
+    const float* __restrict__ slopes, uint16_t* __restrict__ dqkv, int H, int T,
+    int C, float scale, float p_drop, uint32_t seed) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint16_t* k_lds = (uint16_t*)smem;
   uint16_t* v_lds = k_lds + TB * 128;
 
   const int bh = blockIdx.x;  // grid: (BH, tiles) for per-CU load balance
   const int h = bh % H;
-  const long base = (long)bh * T * D;
+  const int QS = 3 * C;
+  const long base = (long)(bh / H) * T * QS + h * D;   // q plane of qkv
+  const long kbase = base + C;
+  const long vbase = base + 2 * C;
+  const long dobase = (long)(bh / H) * T * C + h * D;  // dout plane
   const int q0 = blockIdx.y * RB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -97,8 +107,8 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
 #pragma unroll
     for (int s = 0; s < KS; ++s) {
       if (ok) {
-        q_frag[s] = *reinterpret_cast<const bf16x8*>(&q[base + (long)qi * D + s * 16 + 8 * hi]);
-        do_frag[s] = *reinterpret_cast<const bf16x8*>(&dout[base + (long)qi * D + s * 16 + 8 * hi]);
+        q_frag[s] = *reinterpret_cast<const bf16x8*>(&qkv[base + (long)qi * QS + s * 16 + 8 * hi]);
+        do_frag[s] = *reinterpret_cast<const bf16x8*>(&dout[dobase + (long)qi * C + s * 16 + 8 * hi]);
       } else {
         q_frag[s] = bf16x8{};
         do_frag[s] = bf16x8{};
@@ -116,8 +126,8 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
     __builtin_amdgcn_s_setprio(1);  // static priority for the younger half (T5)
 
   Stage<D> sk, sv;
-  sk.load(k, base, 0, T);
-  sv.load(v, base, 0, T);
+  sk.load(qkv, kbase, QS, 0, T);
+  sv.load(qkv, vbase, QS, 0, T);
   const int kv_end = min(T, q0 + RB);
   for (int kt = 0; kt < kv_end; kt += TB) {
     __syncthreads();  // previous tile fully consumed
@@ -125,8 +135,8 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
     sv.store(v_lds);
     __syncthreads();  // tile kt visible
     if (kt + TB < kv_end) {
-      sk.load(k, base, kt + TB, T);
-      sv.load(v, base, kt + TB, T);
+      sk.load(qkv, kbase, QS, kt + TB, T);
+      sv.load(qkv, vbase, QS, kt + TB, T);
     }
 
 #pragma unroll
@@ -197,7 +207,7 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
     if (qr >= T) continue;
 #pragma unroll
     for (int d = 0; d < DB; ++d)
-      dq[base + (long)qr * D + d * 32 + li] = f32_to_bf16(dq_acc[d][r]);
+      dqkv[base + (long)qr * QS + d * 32 + li] = f32_to_bf16(dq_acc[d][r]);
   }
 }
 
@@ -207,12 +217,10 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
 // ---------------------------------------------------------------------------
 template <int D>
 __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
-    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
-    const uint16_t* __restrict__ v, const uint16_t* __restrict__ dout,
+    const uint16_t* __restrict__ qkv, const uint16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    const float* __restrict__ slopes, uint16_t* __restrict__ dk,
-    uint16_t* __restrict__ dv, int H, int T, float scale, float p_drop,
-    uint32_t seed) {
+    const float* __restrict__ slopes, uint16_t* __restrict__ dqkv, int H, int T,
+    int C, float scale, float p_drop, uint32_t seed) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint16_t* q_lds = (uint16_t*)smem;
   uint16_t* do_lds = q_lds + TB * 128;
@@ -221,7 +229,11 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
 
   const int bh = blockIdx.x;  // grid: (BH, tiles) for per-CU load balance
   const int h = bh % H;
-  const long base = (long)bh * T * D;
+  const int QS = 3 * C;
+  const long base = (long)(bh / H) * T * QS + h * D;   // q plane of qkv
+  const long kbase = base + C;
+  const long vbase = base + 2 * C;
+  const long dobase = (long)(bh / H) * T * C + h * D;
   const int k0 = blockIdx.y * RB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -241,8 +253,8 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
 #pragma unroll
     for (int s = 0; s < KS; ++s) {
       if (ok) {
-        k_frag[s] = *reinterpret_cast<const bf16x8*>(&k[base + (long)kj * D + s * 16 + 8 * hi]);
-        v_frag[s] = *reinterpret_cast<const bf16x8*>(&v[base + (long)kj * D + s * 16 + 8 * hi]);
+        k_frag[s] = *reinterpret_cast<const bf16x8*>(&qkv[kbase + (long)kj * QS + s * 16 + 8 * hi]);
+        v_frag[s] = *reinterpret_cast<const bf16x8*>(&qkv[vbase + (long)kj * QS + s * 16 + 8 * hi]);
       } else {
         k_frag[s] = bf16x8{};
         v_frag[s] = bf16x8{};
@@ -271,8 +283,8 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
     }
   };
 
-  sq.load(q, base, k0, T);
-  sdo.load(dout, base, k0, T);
+  sq.load(qkv, base, QS, k0, T);
+  sdo.load(dout, dobase, C, k0, T);
   load_stats(k0);
   for (int qt = k0; qt < T; qt += TB) {
     __syncthreads();
@@ -284,8 +296,8 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
     }
     __syncthreads();
     if (qt + TB < T) {
-      sq.load(q, base, qt + TB, T);
-      sdo.load(dout, base, qt + TB, T);
+      sq.load(qkv, base, QS, qt + TB, T);
+      sdo.load(dout, dobase, C, qt + TB, T);
       load_stats(qt + TB);
     }
 
@@ -354,46 +366,44 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
     if (kr >= T) continue;
 #pragma unroll
     for (int d = 0; d < DB; ++d) {
-      dk[base + (long)kr * D + d * 32 + li] = f32_to_bf16(dk_acc[d][r]);
-      dv[base + (long)kr * D + d * 32 + li] = f32_to_bf16(dv_acc[d][r]);
+      dqkv[kbase + (long)kr * QS + d * 32 + li] = f32_to_bf16(dk_acc[d][r]);
+      dqkv[vbase + (long)kr * QS + d * 32 + li] = f32_to_bf16(dv_acc[d][r]);
     }
   }
 }
 
 template <int D>
-void launch_bwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
-                const at::Tensor& dout, const at::Tensor& lse, const at::Tensor& delta,
-                const at::Tensor& slopes, at::Tensor& dq, at::Tensor& dk,
-                at::Tensor& dv, int B, int H, int T, float scale, float p_drop,
+void launch_bwd(const at::Tensor& qkv, const at::Tensor& dout, const at::Tensor& lse,
+                const at::Tensor& delta, const at::Tensor& slopes, at::Tensor& dqkv,
+                int B, int H, int T, int C, float scale, float p_drop,
                 uint32_t seed, hipStream_t stream) {
   dim3 grid(B * H, (T + RB - 1) / RB);
   const size_t smem_dq = 2 * TB * 128 * sizeof(uint16_t);
   hipLaunchKernelGGL(flash_dq_kernel<D>, grid, dim3(512), smem_dq, stream,
-                     (const uint16_t*)q.data_ptr(), (const uint16_t*)k.data_ptr(),
-                     (const uint16_t*)v.data_ptr(), (const uint16_t*)dout.data_ptr(),
+                     (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     slopes.data_ptr<float>(), (uint16_t*)dq.data_ptr(), H, T, scale,
-                     p_drop, seed);
+                     slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,
+                     scale, p_drop, seed);
   const size_t smem_kv = 2 * TB * 128 * sizeof(uint16_t) + 2 * TB * sizeof(float);
   hipLaunchKernelGGL(flash_dkdv_kernel<D>, grid, dim3(512), smem_kv, stream,
-                     (const uint16_t*)q.data_ptr(), (const uint16_t*)k.data_ptr(),
-                     (const uint16_t*)v.data_ptr(), (const uint16_t*)dout.data_ptr(),
+                     (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     slopes.data_ptr<float>(), (uint16_t*)dk.data_ptr(),
-                     (uint16_t*)dv.data_ptr(), H, T, scale, p_drop, seed);
+                     slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,
+                     scale, p_drop, seed);
 }
 
 }  // namespace
 
-std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
-                                 at::Tensor v, at::Tensor slopes, at::Tensor o,
-                                 at::Tensor lse, double p_drop, int64_t seed) {
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.dim() == 4);
-  const int B = q.size(0), H = q.size(1), T = q.size(2), D = q.size(3);
-  auto dq = at::empty_like(q);
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
-  auto delta = at::empty({B, H, T}, q.options().dtype(at::kFloat));
+std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor qkv, at::Tensor slopes,
+                                 at::Tensor o, at::Tensor lse, int64_t H_,
+                                 double p_drop, int64_t seed) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 3);
+  TORCH_CHECK(dout.is_contiguous() && dout.dim() == 3);
+  const int B = qkv.size(0), T = qkv.size(1), H = (int)H_;
+  const int C = qkv.size(2) / 3;
+  const int D = C / H;
+  auto dqkv = at::empty_like(qkv);
+  auto delta = at::empty({B, H, T}, qkv.options().dtype(at::kFloat));
   auto sl = slopes.to(at::kFloat).contiguous();
   const float scale = 1.0f / sqrtf((float)D);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
@@ -402,14 +412,15 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
     const int block = 256;
     hipLaunchKernelGGL(delta_kernel, dim3(capped_grid(rows * WAVE, block)), dim3(block),
                        0, stream, (const uint16_t*)dout.data_ptr(),
-                       (const uint16_t*)o.data_ptr(), delta.data_ptr<float>(), rows, D);
+                       (const uint16_t*)o.data_ptr(), delta.data_ptr<float>(), rows,
+                       H, T, C, D);
   }
   switch (D) {
-    case 32: launch_bwd<32>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
-    case 64: launch_bwd<64>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
-    case 96: launch_bwd<96>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
-    case 128: launch_bwd<128>(q, k, v, dout, lse, delta, sl, dq, dk, dv, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 32: launch_bwd<32>(qkv, dout, lse, delta, sl, dqkv, B, H, T, C, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 64: launch_bwd<64>(qkv, dout, lse, delta, sl, dqkv, B, H, T, C, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 96: launch_bwd<96>(qkv, dout, lse, delta, sl, dqkv, B, H, T, C, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 128: launch_bwd<128>(qkv, dout, lse, delta, sl, dqkv, B, H, T, C, scale, (float)p_drop, (uint32_t)seed, stream); break;
     default: TORCH_CHECK(false, "attn_bwd: head_dim must be 32/64/96/128, got ", D);
   }
-  return {dq, dk, dv};
+  return {dqkv};
 }

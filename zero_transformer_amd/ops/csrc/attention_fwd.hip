@@ -44,11 +44,14 @@ using attn::tr_frag;
 
 constexpr float NEG_INF = -3.0e38f;
 
+// qkv is the fused projection output (B, T, 3C) with C = H*D (q at column
+// offset h*D, k at C + h*D, v at 2C + h*D); o is (B, T, C). This is the
+// natural layout of the model's single qkv GEMM — no transposes or
+// .contiguous() copies on either side of the op.
 template <int D>
 __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
-    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
-    const uint16_t* __restrict__ v, const float* __restrict__ slopes,
-    uint16_t* __restrict__ o, float* __restrict__ lse, int H, int T,
+    const uint16_t* __restrict__ qkv, const float* __restrict__ slopes,
+    uint16_t* __restrict__ o, float* __restrict__ lse, int H, int T, int C,
     float scale, float p_drop, uint32_t seed) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint16_t* k_lds = (uint16_t*)smem;       // 64 x [256 B], swizzled
@@ -57,7 +60,11 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
   const int bh = blockIdx.x;  // grid: (BH, tiles) — consecutive blocks share the
   // tile index so per-CU work is balanced across the causal triangle
   const int h = bh % H;
-  const long base = (long)bh * T * D;
+  const int QS = 3 * C;  // qkv row stride (elements)
+  const long base = (long)(bh / H) * T * QS + h * D;   // q plane
+  const long kbase = base + C;
+  const long vbase = base + 2 * C;
+  const long obase = (long)(bh / H) * T * C + h * D;   // o plane
   const int q0 = blockIdx.y * RB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -80,7 +87,7 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
 #pragma unroll
     for (int s = 0; s < KS; ++s)
       q_frag[s] = ok ? *reinterpret_cast<const bf16x8*>(
-                           &q[base + (long)qi * D + s * 16 + 8 * hi])
+                           &qkv[base + (long)qi * QS + s * 16 + 8 * hi])
                      : bf16x8{};
   }
 
@@ -93,8 +100,8 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
     __builtin_amdgcn_s_setprio(1);  // static priority for the younger half (T5)
 
   Stage<D> sk, sv;
-  sk.load(k, base, 0, T);
-  sv.load(v, base, 0, T);
+  sk.load(qkv, kbase, QS, 0, T);
+  sv.load(qkv, vbase, QS, 0, T);
   const int kv_end = min(T, q0 + RB);
   for (int kt = 0; kt < kv_end; kt += TB) {
     __syncthreads();  // previous tile fully consumed
@@ -102,8 +109,8 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
     sv.store(v_lds);
     __syncthreads();  // tile kt visible
     if (kt + TB < kv_end) {
-      sk.load(k, base, kt + TB, T);
-      sv.load(v, base, kt + TB, T);
+      sk.load(qkv, kbase, QS, kt + TB, T);
+      sv.load(qkv, vbase, QS, kt + TB, T);
     }
 
 #pragma unroll
@@ -204,41 +211,43 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
     const float il = __shfl(inv_l, row, 64);
 #pragma unroll
     for (int d = 0; d < DB; ++d)
-      o[base + (long)qr * D + d * 32 + li] = f32_to_bf16(o_acc[d][r] * il);
+      o[obase + (long)qr * C + d * 32 + li] = f32_to_bf16(o_acc[d][r] * il);
   }
 }
 
 template <int D>
-void launch_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
-                const at::Tensor& slopes, at::Tensor& o, at::Tensor& lse, int B,
-                int H, int T, float scale, float p_drop, uint32_t seed,
-                hipStream_t stream) {
+void launch_fwd(const at::Tensor& qkv, const at::Tensor& slopes, at::Tensor& o,
+                at::Tensor& lse, int B, int H, int T, int C, float scale,
+                float p_drop, uint32_t seed, hipStream_t stream) {
   dim3 grid(B * H, (T + RB - 1) / RB);
   const size_t smem = 2 * TB * 128 * sizeof(uint16_t);
   hipLaunchKernelGGL(flash_fwd_kernel<D>, grid, dim3(512), smem, stream,
-                     (const uint16_t*)q.data_ptr(), (const uint16_t*)k.data_ptr(),
-                     (const uint16_t*)v.data_ptr(), slopes.data_ptr<float>(),
-                     (uint16_t*)o.data_ptr(), lse.data_ptr<float>(), H, T, scale,
-                     p_drop, seed);
+                     (const uint16_t*)qkv.data_ptr(), slopes.data_ptr<float>(),
+                     (uint16_t*)o.data_ptr(), lse.data_ptr<float>(), H, T, C,
+                     scale, p_drop, seed);
 }
 
 }  // namespace
 
-std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
-                                 at::Tensor slopes, double p_drop, int64_t seed) {
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.dim() == 4, "q must be (B,H,T,D) contiguous");
-  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_fwd: bf16 only");
-  const int B = q.size(0), H = q.size(1), T = q.size(2), D = q.size(3);
-  auto o = at::empty_like(q);
-  auto lse = at::empty({B, H, T}, q.options().dtype(at::kFloat));
+std::vector<at::Tensor> attn_fwd(at::Tensor qkv, at::Tensor slopes, int64_t H_,
+                                 double p_drop, int64_t seed) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 3,
+              "qkv must be (B, T, 3C) contiguous");
+  TORCH_CHECK(qkv.scalar_type() == at::kBFloat16, "attn_fwd: bf16 only");
+  const int B = qkv.size(0), T = qkv.size(1), H = (int)H_;
+  const int C = qkv.size(2) / 3;
+  TORCH_CHECK(qkv.size(2) == 3 * C && C % H == 0, "bad qkv shape");
+  const int D = C / H;
+  auto o = at::empty({B, T, C}, qkv.options());
+  auto lse = at::empty({B, H, T}, qkv.options().dtype(at::kFloat));
   auto sl = slopes.to(at::kFloat).contiguous();
   const float scale = 1.0f / sqrtf((float)D);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   switch (D) {
-    case 32: launch_fwd<32>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
-    case 64: launch_fwd<64>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
-    case 96: launch_fwd<96>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
-    case 128: launch_fwd<128>(q, k, v, sl, o, lse, B, H, T, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 32: launch_fwd<32>(qkv, sl, o, lse, B, H, T, C, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 64: launch_fwd<64>(qkv, sl, o, lse, B, H, T, C, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 96: launch_fwd<96>(qkv, sl, o, lse, B, H, T, C, scale, (float)p_drop, (uint32_t)seed, stream); break;
+    case 128: launch_fwd<128>(qkv, sl, o, lse, B, H, T, C, scale, (float)p_drop, (uint32_t)seed, stream); break;
     default: TORCH_CHECK(false, "attn_fwd: head_dim must be one of 32/64/96/128, got ", D);
   }
   return {o, lse};

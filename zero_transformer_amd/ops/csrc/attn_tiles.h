@@ -123,14 +123,16 @@ template <int D>
 struct Stage {
   static constexpr int NC = (TB * D + 512 * 8 - 1) / (512 * 8);
   s16x8 r[NC];
-  ZTA_DEV void load(const uint16_t* g, long base, int row0, int T) {
+  // `base` points at row 0 of this (b, h) plane; `rs` is the row stride in
+  // elements (3*C for tensors packed as (B, T, 3C) qkv, C for (B, T, C)).
+  ZTA_DEV void load(const uint16_t* g, long base, int rs, int row0, int T) {
     const int t = threadIdx.x;
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
       const int idx = t * 8 + c * 512 * 8;
       const int rg = row0 + idx / D;
       r[c] = (idx < TB * D && rg < T)
-                 ? *reinterpret_cast<const s16x8*>(&g[base + (long)rg * D + idx % D])
+                 ? *reinterpret_cast<const s16x8*>(&g[base + (long)rg * rs + idx % D])
                  : s16x8{};
     }
   }

@@ -15,11 +15,11 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor l
 void adamw_step(at::Tensor p, at::Tensor p_bf16, at::Tensor g, at::Tensor m,
                 at::Tensor v, long step, double lr, double beta1, double beta2,
                 double eps, double wd, double clip, double grad_scale);
-std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
-                                 at::Tensor slopes, double p_drop, int64_t seed);
-std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
-                                 at::Tensor v, at::Tensor slopes, at::Tensor o,
-                                 at::Tensor lse, double p_drop, int64_t seed);
+std::vector<at::Tensor> attn_fwd(at::Tensor qkv, at::Tensor slopes, int64_t H,
+                                 double p_drop, int64_t seed);
+std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor qkv, at::Tensor slopes,
+                                 at::Tensor o, at::Tensor lse, int64_t H,
+                                 double p_drop, int64_t seed);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "bias-free LayerNorm fwd (gfx950)");

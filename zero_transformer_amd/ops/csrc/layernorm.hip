@@ -129,13 +129,18 @@ __global__ __launch_bounds__(1024) void ln_bwd_vec(
   *reinterpret_cast<f32x4*>(&dw_part[(long)blockIdx.x * C + t * 8 + 4]) = p1;
 }
 
+// grid (C/256, NSPLIT): y-block sums its slice of partial rows, one
+// atomicAdd per column per slice (C*NSPLIT atomics total — negligible).
 __global__ void ln_dw_reduce(const float* __restrict__ dw_part,
                              float* __restrict__ dw, int C, int npart) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  const int per = (npart + gridDim.y - 1) / gridDim.y;
+  const int b0 = blockIdx.y * per;
+  const int b1 = min(npart, b0 + per);
   float s = 0.f;
-  for (int b = 0; b < npart; ++b) s += dw_part[(long)b * C + c];
-  dw[c] = s;
+  for (int b = b0; b < b1; ++b) s += dw_part[(long)b * C + c];
+  atomicAdd(&dw[c], s);
 }
 
 // ---------------------------------------------------------------------------
@@ -263,7 +268,7 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
                        (const uint16_t*)w.data_ptr(), mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), (uint16_t*)dx.data_ptr(),
                        dw_part.data_ptr<float>(), rows, C);
-    hipLaunchKernelGGL(ln_dw_reduce, dim3((C + 255) / 256), dim3(256), 0, stream,
+    hipLaunchKernelGGL(ln_dw_reduce, dim3((C + 255) / 256, 16), dim3(256), 0, stream,
                        dw_part.data_ptr<float>(), dw_f32.data_ptr<float>(), C, grid);
   } else {
     const int block = 256;
