@@ -66,15 +66,16 @@ class CausalSelfAttention(nn.Module):
             self.register_buffer("slopes", ops.alibi_slopes(heads), persistent=False)
         else:
             self.slopes = None
-        self.resid_dropout = nn.Dropout(cfg.dropout)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        """Returns the PRE-dropout attention output; the residual dropout is
+        fused into the Block's residual add (ops.residual_dropout_add)."""
         qkv = F.linear(x, self.qkv_w)  # (B, T, 3C)
         o = ops.attention_qkv(
             qkv, self.num_head, self.slopes,
             dropout_p=self.dropout_p, training=self.training,
         )
-        return self.resid_dropout(self.fc_resid(o))
+        return self.fc_resid(o)
 
     # .pth contract: expose query/key/value instead of the fused qkv_w
     def _save_to_state_dict(self, destination, prefix, keep_vars):
@@ -96,30 +97,37 @@ class CausalSelfAttention(nn.Module):
 
 
 class MLP(nn.Module):
-    """4x expansion GELU MLP (reference layers.py:47-77)."""
+    """4x expansion GELU MLP (reference layers.py:47-77). The trailing
+    dropout is fused into the Block's residual add."""
 
     def __init__(self, cfg):
         super().__init__()
         dim = cfg.embedding_dim
         self.fc1 = nn.Linear(dim, 4 * dim, bias=False)
         self.fc_resid = nn.Linear(4 * dim, dim, bias=False)
-        self.dropout = nn.Dropout(cfg.dropout)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.dropout(self.fc_resid(ops.gelu(self.fc1(x))))
+        return self.fc_resid(ops.gelu(self.fc1(x)))
 
 
 class Block(nn.Module):
     def __init__(self, cfg):
         super().__init__()
+        self.dropout_p = cfg.dropout
         self.ln1 = LayerNorm(cfg.embedding_dim)
         self.attn = CausalSelfAttention(cfg)
         self.ln2 = LayerNorm(cfg.embedding_dim)
         self.mlp = MLP(cfg)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = x + self.attn(self.ln1(x))
-        x = x + self.mlp(self.ln2(x))
+        # residual adds with the reference's resid/MLP dropout fused in
+        # (dropout(h) then add, layers.py:76,190 + GPT.py:43-49)
+        x = ops.residual_dropout_add(
+            x, self.attn(self.ln1(x)), self.dropout_p, self.training
+        )
+        x = ops.residual_dropout_add(
+            x, self.mlp(self.ln2(x)), self.dropout_p, self.training
+        )
         return x
 
 

@@ -203,6 +203,41 @@ def attention(
 
 
 # ---------------------------------------------------------------------------
+# Fused residual-add + dropout:  y = x + dropout(h, p)
+# ---------------------------------------------------------------------------
+
+
+class _ResidualDropoutFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, h, p, training):
+        ext = hip_ops()
+        if p > 0.0 and training:
+            seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+        else:
+            seed, p = 0, 0.0
+        ctx.p = p
+        ctx.seed = seed
+        return ext.residual_dropout_fwd(x.contiguous(), h.contiguous(), float(p), seed)
+
+    @staticmethod
+    def backward(ctx, dy):
+        dy = dy.contiguous()
+        if ctx.p == 0.0:
+            return dy, dy, None, None
+        dh = hip_ops().residual_dropout_bwd(dy, ctx.p, ctx.seed)
+        return dy, dh, None, None
+
+
+def residual_dropout_add(
+    x: torch.Tensor, h: torch.Tensor, p: float, training: bool
+) -> torch.Tensor:
+    """y = x + dropout(h, p) in one memory pass (mask regenerated in bwd)."""
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.numel() % 8 == 0:
+        return _ResidualDropoutFn.apply(x, h, p, training)
+    return x + torch.nn.functional.dropout(h, p=p, training=training)
+
+
+# ---------------------------------------------------------------------------
 # Fused cross entropy (gather-based, never one-hot)
 # ---------------------------------------------------------------------------
 

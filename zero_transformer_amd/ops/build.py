@@ -26,6 +26,7 @@ SOURCES = [
     "gelu.hip",
     "cross_entropy.hip",
     "adamw.hip",
+    "residual.hip",
     "attention_fwd.hip",
     "attention_bwd.hip",
 ]

@@ -15,6 +15,8 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor l
 void adamw_step(at::Tensor p, at::Tensor p_bf16, at::Tensor g, at::Tensor m,
                 at::Tensor v, long step, double lr, double beta1, double beta2,
                 double eps, double wd, double clip, double grad_scale);
+at::Tensor residual_dropout_fwd(at::Tensor x, at::Tensor h, double p, int64_t seed);
+at::Tensor residual_dropout_bwd(at::Tensor dy, double p, int64_t seed);
 std::vector<at::Tensor> attn_fwd(at::Tensor qkv, at::Tensor slopes, int64_t H,
                                  double p_drop, int64_t seed);
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor qkv, at::Tensor slopes,
@@ -29,6 +31,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused gather CE fwd (gfx950)");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused gather CE bwd (gfx950)");
   m.def("adamw_step", &adamw_step, "fused ZeRO-1 AdamW shard step (gfx950)");
+  m.def("residual_dropout_fwd", &residual_dropout_fwd, "fused x + dropout(h) fwd (gfx950)");
+  m.def("residual_dropout_bwd", &residual_dropout_bwd, "fused x + dropout(h) bwd (gfx950)");
   m.def("attn_fwd", &attn_fwd, "fused causal ALiBi flash attention fwd (gfx950 MFMA)");
   m.def("attn_bwd", &attn_bwd, "fused causal ALiBi flash attention bwd (gfx950 MFMA)");
 }
