@@ -210,7 +210,6 @@ class ZeRO1Optimizer:
                 self._launch_reduce(b)
         if self.overlap_comm:
             torch.cuda.current_stream(self.device).wait_stream(self._comm_stream)
-        shard_n = None
         for b in self.buckets:
             shard_n = b.numel // self.world
             own = b.flat_param[self.rank * shard_n : (self.rank + 1) * shard_n]
@@ -231,7 +230,18 @@ class ZeRO1Optimizer:
             )
             if self.param_dtype != torch.bfloat16:
                 own.copy_(b.master.to(self.param_dtype))
-            comm.all_gather_flat(b.flat_param, own)
+            if self.world > 1:
+                # pipeline: gather bucket i on the comm stream while bucket
+                # i+1's AdamW runs on the compute stream
+                if self.overlap_comm:
+                    self._comm_stream.wait_stream(torch.cuda.current_stream(self.device))
+                    with torch.cuda.stream(self._comm_stream):
+                        comm.all_gather_flat(b.flat_param, own)
+                else:
+                    comm.all_gather_flat(b.flat_param, own)
+            # world == 1: `own` aliases the full flat_param — nothing to gather
+        if self.overlap_comm:
+            torch.cuda.current_stream(self.device).wait_stream(self._comm_stream)
         self.zero_grad()
         return lr
 
