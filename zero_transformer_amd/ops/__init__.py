@@ -244,23 +244,29 @@ def residual_dropout_add(
 
 class _CrossEntropyFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, logits, targets):
+    def forward(ctx, logits, targets, divisor):
         logits = logits.contiguous()
-        loss, lse = hip_ops().cross_entropy_fwd(logits, targets)
+        loss, lse = hip_ops().cross_entropy_fwd(logits, targets, divisor)
         ctx.save_for_backward(logits, targets, lse)
+        ctx.divisor = divisor
         return loss
 
     @staticmethod
     def backward(ctx, dloss):
         logits, targets, lse = ctx.saved_tensors
-        dlogits = hip_ops().cross_entropy_bwd(logits, targets, lse, dloss.contiguous())
-        return dlogits, None
+        dlogits = hip_ops().cross_entropy_bwd(
+            logits, targets, lse, dloss.contiguous(), ctx.divisor
+        )
+        return dlogits, None, None
 
 
-def cross_entropy(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
-    """Mean CE over rows; logits (N, V) any float dtype, targets (N,) int64."""
+def cross_entropy(
+    logits: torch.Tensor, targets: torch.Tensor, divisor: int = 0
+) -> torch.Tensor:
+    """Mean CE; logits (N, V), targets (N,) int64 (rows with target -1 are
+    ignored). `divisor` = number of counted rows (0: all N rows)."""
     if logits.is_cuda:
-        return _CrossEntropyFn.apply(logits, targets)
+        return _CrossEntropyFn.apply(logits, targets, divisor)
     return reference.cross_entropy(logits, targets)
 
 

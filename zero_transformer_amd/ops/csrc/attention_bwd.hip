@@ -134,6 +134,7 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
     sk.store(k_lds);
     sv.store(v_lds);
     __syncthreads();  // tile kt visible
+    attn::lds_acquire();
     if (kt + TB < kv_end) {
       sk.load(qkv, kbase, QS, kt + TB, T);
       sv.load(qkv, vbase, QS, kt + TB, T);
@@ -295,11 +296,14 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
       delta_s[t512] = delta_reg;
     }
     __syncthreads();
+    attn::lds_acquire();
+#ifndef ZTA_DKDV_NO_STAGE  // perf ablation: expose staging cost
     if (qt + TB < T) {
       sq.load(qkv, base, QS, qt + TB, T);
       sdo.load(dout, dobase, C, qt + TB, T);
       load_stats(qt + TB);
     }
+#endif
 
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
@@ -345,16 +349,18 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
       c_to_a_frags(p_pv, pa);  // A[i=key][k=q]
       c_to_a_frags(ds, dsa);
 
-      // dV += P^T dO ; dK += dS^T Q : B[k=q][j=d] via tr-reads
+      // dV += P^T dO ; dK += dS^T Q : B[k=q][j=d] via tr-reads (all eight
+      // transpose reads of a d-block in one asm, one lgkmcnt drain)
+#ifndef ZTA_DKDV_NO_ACCUM  // perf ablation: expose accumulation-loop cost
 #pragma unroll
       for (int d = 0; d < DB; ++d) {
-        attn::TrPair dop = attn::tr_frag_pair(do_lds, sub * 32, d * 32);
-        dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], dop.a, dv_acc[d], 0, 0, 0);
-        dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], dop.b, dv_acc[d], 0, 0, 0);
-        attn::TrPair qp = attn::tr_frag_pair(q_lds, sub * 32, d * 32);
-        dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[0], qp.a, dk_acc[d], 0, 0, 0);
-        dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[1], qp.b, dk_acc[d], 0, 0, 0);
+        attn::TrQuad t = attn::tr_frag_quad(do_lds, q_lds, sub * 32, d * 32);
+        dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], t.x.a, dv_acc[d], 0, 0, 0);
+        dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], t.x.b, dv_acc[d], 0, 0, 0);
+        dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[0], t.y.a, dk_acc[d], 0, 0, 0);
+        dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[1], t.y.b, dk_acc[d], 0, 0, 0);
       }
+#endif
     }
   }
 

@@ -108,6 +108,7 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
     sk.store(k_lds);
     sv.store(v_lds);
     __syncthreads();  // tile kt visible
+    attn::lds_acquire();
     if (kt + TB < kv_end) {
       sk.load(qkv, kbase, QS, kt + TB, T);
       sv.load(qkv, vbase, QS, kt + TB, T);

@@ -25,15 +25,22 @@ ZTA_DEV int swz(int row, int byte_off) { return byte_off ^ ((row & 7) << 4); }
 // each 16-lane group, out[lane 4a+b][reg j] = in[lane 4j+a][elem b] — i.e.
 // lane l supplies row ((l>>2)&3), column-block 4*(l&3) of a [4][16] tile and
 // receives the column (l%16) of that tile, rows ascending over the 4 regs.
-// NOTE: reads + waitcnt MUST be one asm statement. SIInsertWaitcnts cannot
-// see a ds_read inside inline asm, so it inserts no lgkmcnt wait before uses
-// of its outputs, and a separate waitcnt asm has no dataflow edge to the
-// outputs — the scheduler may move the consuming MFMA between read and wait
-// (observed: register-junk O values with exact lse). The "memory" clobber is
-// equally required: without it the compiler neither orders other waves'
-// staging stores against the asm read nor keeps them alive (no visible
-// reader). "=&v" keeps destinations from aliasing still-live address
-// operands.
+// Ordering contract for the transpose-read asms below:
+//  * reads + waitcnt MUST be one asm statement: SIInsertWaitcnts cannot see
+//    a ds_read inside inline asm, so it inserts no lgkmcnt wait before uses
+//    of its outputs, and a separate waitcnt asm has no dataflow edge to the
+//    outputs — the scheduler may move the consuming MFMA between read and
+//    wait (observed: register-junk O values with exact lse).
+//  * the asms carry NO "memory" clobber — a clobber on every fragment read
+//    made each one a full scheduling barrier and serialized the whole MFMA
+//    accumulation loop. Instead the kernel calls lds_acquire() ONCE after
+//    the tile-ready __syncthreads(): volatile asms are not reordered with
+//    respect to each other, so the clobbered empty asm (a) keeps the
+//    staging stores alive (a may-read-everything point after them) and
+//    (b) pins every later volatile tr-read below the barrier.
+ZTA_DEV void lds_acquire() { asm volatile("" ::: "memory"); }
+
+// "=&v" keeps destinations from aliasing still-live address operands.
 ZTA_DEV bf16x8 tr_frag(const uint16_t* lds, int k0, int j0) {
   const int l = threadIdx.x & 63;
   const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
@@ -50,8 +57,7 @@ ZTA_DEV bf16x8 tr_frag(const uint16_t* lds, int k0, int j0) {
       "ds_read_b64_tr_b16 %1, %3\n\t"
       "s_waitcnt lgkmcnt(0)"
       : "=&v"(u.d[0]), "=&v"(u.d[1])
-      : "v"(a0), "v"(a1)
-      : "memory");
+      : "v"(a0), "v"(a1));
   return u.v;
 }
 
@@ -61,6 +67,45 @@ ZTA_DEV bf16x8 tr_frag(const uint16_t* lds, int k0, int j0) {
 struct TrPair {
   bf16x8 a, b;
 };
+
+// Pairs from TWO tiles (same k0/j0) in one asm: eight transpose reads, one
+// drain — for the dKdV loop, which needs dO^T and Q^T fragments per d.
+struct TrQuad {
+  TrPair x, y;
+};
+ZTA_DEV TrQuad tr_frag_quad(const uint16_t* lds_x, const uint16_t* lds_y, int k0,
+                            int j0) {
+  const int l = threadIdx.x & 63;
+  const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
+  const int rb = 8 * (l >> 5) + ((l >> 2) & 3);
+  int ax[4], ay[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int row = k0 + 16 * (i >> 1) + 4 * (i & 1) + rb;
+    const int off = row * 256 + (colb ^ ((row & 7) << 4));
+    ax[i] = (int)(size_t)((const char*)lds_x + off);
+    ay[i] = (int)(size_t)((const char*)lds_y + off);
+  }
+  union {
+    i32x2 d[8];
+    TrQuad f;
+  } u;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %8\n\t"
+      "ds_read_b64_tr_b16 %1, %9\n\t"
+      "ds_read_b64_tr_b16 %2, %10\n\t"
+      "ds_read_b64_tr_b16 %3, %11\n\t"
+      "ds_read_b64_tr_b16 %4, %12\n\t"
+      "ds_read_b64_tr_b16 %5, %13\n\t"
+      "ds_read_b64_tr_b16 %6, %14\n\t"
+      "ds_read_b64_tr_b16 %7, %15\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(u.d[0]), "=&v"(u.d[1]), "=&v"(u.d[2]), "=&v"(u.d[3]),
+        "=&v"(u.d[4]), "=&v"(u.d[5]), "=&v"(u.d[6]), "=&v"(u.d[7])
+      : "v"(ax[0]), "v"(ax[1]), "v"(ax[2]), "v"(ax[3]), "v"(ay[0]), "v"(ay[1]),
+        "v"(ay[2]), "v"(ay[3]));
+  return u.f;
+}
 ZTA_DEV TrPair tr_frag_pair(const uint16_t* lds, int k0, int j0) {
   const int l = threadIdx.x & 63;
   const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
@@ -82,8 +127,7 @@ ZTA_DEV TrPair tr_frag_pair(const uint16_t* lds, int k0, int j0) {
       "ds_read_b64_tr_b16 %3, %7\n\t"
       "s_waitcnt lgkmcnt(0)"
       : "=&v"(u.d[0]), "=&v"(u.d[1]), "=&v"(u.d[2]), "=&v"(u.d[3])
-      : "v"(a[0]), "v"(a[1]), "v"(a[2]), "v"(a[3])
-      : "memory");
+      : "v"(a[0]), "v"(a[1]), "v"(a[2]), "v"(a[3]));
   return u.f;
 }
 

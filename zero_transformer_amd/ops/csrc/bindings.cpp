@@ -9,9 +9,10 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
                                       at::Tensor rstd, at::Tensor mean);
 at::Tensor gelu_fwd(at::Tensor x);
 at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x);
-std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets);
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets,
+                                          int64_t divisor);
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
-                             at::Tensor dloss);
+                             at::Tensor dloss, int64_t divisor);
 void adamw_step(at::Tensor p, at::Tensor p_bf16, at::Tensor g, at::Tensor m,
                 at::Tensor v, long step, double lr, double beta1, double beta2,
                 double eps, double wd, double clip, double grad_scale);

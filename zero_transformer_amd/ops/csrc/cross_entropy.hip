@@ -66,7 +66,8 @@ __global__ __launch_bounds__(256) void ce_fwd_vec(
     const float lse = m + __logf(s);
     if (t == 0) {
       lse_out[row] = lse;
-      loss[row] = lse - bf16_to_f32(lr[targets[row]]);
+      const long tg = targets[row];
+      loss[row] = tg >= 0 ? lse - bf16_to_f32(lr[tg]) : 0.f;
     }
   }
 }
@@ -74,14 +75,19 @@ __global__ __launch_bounds__(256) void ce_fwd_vec(
 __global__ __launch_bounds__(256) void ce_bwd_vec(
     const uint16_t* __restrict__ logits, const long* __restrict__ targets,
     const float* __restrict__ lse, const float* __restrict__ dloss,
-    uint16_t* __restrict__ dlogits, long rows, int V) {
-  const float scale = dloss[0] / rows;
+    uint16_t* __restrict__ dlogits, long rows, long divisor, int V) {
+  const float scale = dloss[0] / divisor;
   const int t = threadIdx.x;
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const uint16_t* lr = logits + row * V;
     uint16_t* dr = dlogits + row * V;
     const float l = lse[row];
     const long tg = targets[row];
+    if (tg < 0) {  // ignore_index row: zero gradient
+      for (int c = t * 8; c < V; c += 256 * 8)
+        *reinterpret_cast<s16x8*>(&dr[c]) = s16x8{};
+      continue;
+    }
     for (int c = t * 8; c < V; c += 256 * 8) {
       s16x8 x8 = *reinterpret_cast<const s16x8*>(&lr[c]);
       s16x8 d8;
@@ -125,7 +131,8 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
     const float lse = m + logf(s);
     if (threadIdx.x == 0) {
       lse_out[row] = lse;
-      loss[row] = lse - to_f32(lr[targets[row]]);
+      const long tg = targets[row];
+      loss[row] = tg >= 0 ? lse - to_f32(lr[tg]) : 0.f;
     }
     __syncthreads();
   }
@@ -136,13 +143,18 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits,
                               const long* __restrict__ targets,
                               const float* __restrict__ lse,
                               const float* __restrict__ dloss,  // scalar
-                              T* __restrict__ dlogits, long rows, int V) {
-  const float scale = dloss[0] / rows;  // mean over rows
+                              T* __restrict__ dlogits, long rows, long divisor,
+                              int V) {
+  const float scale = dloss[0] / divisor;  // mean over counted rows
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* lr = logits + row * V;
     T* dr = dlogits + row * V;
     const float l = lse[row];
     const long t = targets[row];
+    if (t < 0) {
+      for (int c = threadIdx.x; c < V; c += blockDim.x) dr[c] = from_f32<T>(0.f);
+      continue;
+    }
     for (int c = threadIdx.x; c < V; c += blockDim.x) {
       float p = expf(to_f32(lr[c]) - l);
       dr[c] = from_f32<T>(scale * (p - (c == t ? 1.f : 0.f)));
@@ -152,7 +164,11 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits,
 
 }  // namespace
 
-std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets) {
+// `divisor`: number of rows the mean is taken over (rows with target < 0 are
+// ignored and contribute 0; the caller knows the valid count analytically —
+// the shifted-CE path passes B*(T-1)). divisor <= 0 means all rows count.
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets,
+                                          int64_t divisor) {
   TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() && logits.dim() == 2);
   TORCH_CHECK(targets.scalar_type() == at::kLong);
   const long rows = logits.size(0);
@@ -177,12 +193,14 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets)
   } else {
     TORCH_CHECK(false, "cross_entropy: unsupported dtype");
   }
-  return {loss.mean(), lse};
+  const double div = divisor > 0 ? (double)divisor : (double)rows;
+  return {loss.sum() / div, lse};
 }
 
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
-                             at::Tensor dloss) {
+                             at::Tensor dloss, int64_t divisor) {
   const long rows = logits.size(0);
+  const long div = divisor > 0 ? divisor : rows;
   const int V = logits.size(1);
   auto dlogits = at::empty_like(logits);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
@@ -193,17 +211,17 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor l
     hipLaunchKernelGGL(ce_bwd_vec, dim3(grid), dim3(block), 0, stream,
                        (const uint16_t*)logits.data_ptr(), targets.data_ptr<long>(),
                        lse.data_ptr<float>(), dl.data_ptr<float>(),
-                       (uint16_t*)dlogits.data_ptr(), rows, V);
+                       (uint16_t*)dlogits.data_ptr(), rows, div, V);
   } else if (logits.scalar_type() == at::kBFloat16) {
     hipLaunchKernelGGL(ce_bwd_kernel<uint16_t>, dim3(grid), dim3(block), 0, stream,
                        (const uint16_t*)logits.data_ptr(), targets.data_ptr<long>(),
                        lse.data_ptr<float>(), dl.data_ptr<float>(),
-                       (uint16_t*)dlogits.data_ptr(), rows, V);
+                       (uint16_t*)dlogits.data_ptr(), rows, div, V);
   } else {
     hipLaunchKernelGGL(ce_bwd_kernel<float>, dim3(grid), dim3(block), 0, stream,
                        logits.data_ptr<float>(), targets.data_ptr<long>(),
                        lse.data_ptr<float>(), dl.data_ptr<float>(),
-                       dlogits.data_ptr<float>(), rows, V);
+                       dlogits.data_ptr<float>(), rows, div, V);
   }
   return dlogits;
 }

@@ -161,7 +161,7 @@ def cross_entropy(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
     one-hot * log_softmax mean (src/utils/losses.py:10-23) without
     materializing the one-hot.
     """
-    return F.cross_entropy(logits.float(), targets)
+    return F.cross_entropy(logits.float(), targets, ignore_index=-1)
 
 
 def adamw_update(
