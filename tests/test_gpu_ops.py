@@ -223,6 +223,53 @@ def test_attention_dropout_fwd_bwd_exact_mask_parity(dev, p):
         assert diff / scale < 5e-2, f"{name} rel-max diff {diff/scale} (abs {diff})"
 
 
+def test_residual_dropout_add(dev):
+    """y = x + dropout(h): exact vs the mirrored mask; p=0 is a plain add;
+    backward returns dy and masked dy."""
+    ops = _hip()
+    from zero_transformer_amd import ops as O
+    from zero_transformer_amd.ops import reference
+
+    torch.manual_seed(8)
+    n = 64 * 2048
+    x = torch.randn(n, device=dev).to(torch.bfloat16)
+    h = torch.randn(n, device=dev).to(torch.bfloat16)
+    ext = O.hip_ops()
+    # p = 0: plain fused add
+    y0 = ext.residual_dropout_fwd(x, h, 0.0, 0)
+    assert torch.allclose(y0.float(), (x.float() + h.float()), atol=2e-2)
+    # p = 0.3 with fixed seed: matches the mirrored mask exactly
+    p, seed = 0.3, 777
+    y = ext.residual_dropout_fwd(x, h, p, seed)
+    keep = reference.residual_drop_mask(seed, n, p).to(dev)
+    want = x.float() + h.float() * keep.float() * reference.drop_inv_keep(p)
+    assert (y.float() - want).abs().max().item() < 3e-2
+    # backward: dh = dy * mask * inv_keep
+    dy = torch.randn(n, device=dev).to(torch.bfloat16)
+    dh = ext.residual_dropout_bwd(dy, p, seed)
+    want_dh = dy.float() * keep.float() * reference.drop_inv_keep(p)
+    assert (dh.float() - want_dh).abs().max().item() < 2e-2
+
+
+def test_cross_entropy_ignore_index(dev):
+    """Rows with target -1 contribute 0 loss and 0 grad; mean over divisor."""
+    ops = _hip()
+    torch.manual_seed(9)
+    N, V = 256, 50304
+    logits = (torch.randn(N, V, device=dev) * 2).to(torch.bfloat16).requires_grad_(True)
+    targets = torch.randint(0, V, (N,), device=dev)
+    targets[::4] = -1  # ignore every 4th row
+    divisor = int((targets >= 0).sum().item())
+    loss = ops.cross_entropy(logits, targets, divisor=divisor)
+    lr = logits.detach().float().cpu().requires_grad_(True)
+    lref = torch.nn.functional.cross_entropy(lr, targets.cpu(), ignore_index=-1)
+    assert abs(loss.item() - lref.item()) < 2e-2
+    loss.backward()
+    lref.backward()
+    assert torch.allclose(logits.grad.float().cpu(), lr.grad, atol=1e-4, rtol=5e-2)
+    assert logits.grad[0].abs().max().item() == 0.0  # ignored row: zero grad
+
+
 def test_model_train_step_gpu(dev):
     """End-to-end: one ZeRO-1 train step of a small flagship-shaped model on
     the HIP path; loss finite and decreasing over a few steps."""

@@ -110,6 +110,34 @@ def drop_mask(seed: int, B: int, H: int, T: int, p: float) -> torch.Tensor:
     return torch.from_numpy(keep.reshape(B, H, T, T).copy())
 
 
+def residual_drop_mask(seed: int, n: int, p: float) -> torch.Tensor:
+    """Mirror of csrc/residual.hip's element mask: element i keeps iff byte
+    (i & 3) of drop_bits32(seed, i4 >> 16, i4 & 0xffff) >= thr, i4 = i >> 2."""
+    import numpy as np
+
+    thr = int(p * 256.0 + 0.5)
+
+    def mix32(x):
+        x = x.astype(np.uint32, copy=True)
+        x ^= x >> np.uint32(16)
+        x *= np.uint32(0x85EBCA6B)
+        x ^= x >> np.uint32(13)
+        x *= np.uint32(0xC2B2AE35)
+        x ^= x >> np.uint32(16)
+        return x
+
+    i4 = np.arange((n + 3) // 4, dtype=np.uint64)
+    hi = (i4 >> np.uint64(16)).astype(np.uint32)
+    lo = (i4 & np.uint64(0xFFFF)).astype(np.uint32)
+    with np.errstate(over="ignore"):
+        bits = mix32(
+            np.uint32(seed) ^ (hi * np.uint32(0x9E3779B9)) ^ (lo * np.uint32(0x85EBCA6B))
+        )
+    by = np.stack([(bits >> np.uint32(8 * e)) & np.uint32(0xFF) for e in range(4)], -1)
+    keep = (by >= np.uint32(thr)).reshape(-1)[:n]
+    return torch.from_numpy(keep.copy())
+
+
 def drop_inv_keep(p: float) -> float:
     """Rescale factor matching the kernels' realized 8-bit threshold."""
     thr = int(p * 256.0 + 0.5)
