@@ -378,12 +378,41 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
   }
 }
 
+// dq and dkdv are independent (disjoint column ranges of dqkv) and both
+// stall-bound at one 512-thread block per CU — run them CONCURRENTLY on two
+// streams so their blocks co-reside (227 + 256 VGPR and 32 + 64.5 KB LDS
+// both fit one CU) and each fills the other's wait cycles.
+inline hipStream_t bwd_side_stream() {
+  static hipStream_t s = [] {
+    hipStream_t t;
+    (void)hipStreamCreateWithFlags(&t, hipStreamNonBlocking);
+    return t;
+  }();
+  return s;
+}
+inline hipEvent_t bwd_event(int i) {
+  static hipEvent_t e[2] = {[] {
+                              hipEvent_t t;
+                              (void)hipEventCreateWithFlags(&t, hipEventDisableTiming);
+                              return t;
+                            }(),
+                            [] {
+                              hipEvent_t t;
+                              (void)hipEventCreateWithFlags(&t, hipEventDisableTiming);
+                              return t;
+                            }()};
+  return e[i];
+}
+
 template <int D>
 void launch_bwd(const at::Tensor& qkv, const at::Tensor& dout, const at::Tensor& lse,
                 const at::Tensor& delta, const at::Tensor& slopes, at::Tensor& dqkv,
                 int B, int H, int T, int C, float scale, float p_drop,
                 uint32_t seed, hipStream_t stream) {
   dim3 grid(B * H, (T + RB - 1) / RB);
+  hipStream_t side = bwd_side_stream();
+  (void)hipEventRecord(bwd_event(0), stream);
+  (void)hipStreamWaitEvent(side, bwd_event(0), 0);
   const size_t smem_dq = 2 * TB * 128 * sizeof(uint16_t);
   hipLaunchKernelGGL(flash_dq_kernel<D>, grid, dim3(512), smem_dq, stream,
                      (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
@@ -391,11 +420,13 @@ void launch_bwd(const at::Tensor& qkv, const at::Tensor& dout, const at::Tensor&
                      slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,
                      scale, p_drop, seed);
   const size_t smem_kv = 2 * TB * 128 * sizeof(uint16_t) + 2 * TB * sizeof(float);
-  hipLaunchKernelGGL(flash_dkdv_kernel<D>, grid, dim3(512), smem_kv, stream,
+  hipLaunchKernelGGL(flash_dkdv_kernel<D>, grid, dim3(512), smem_kv, side,
                      (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
                      slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,
                      scale, p_drop, seed);
+  (void)hipEventRecord(bwd_event(1), side);
+  (void)hipStreamWaitEvent(stream, bwd_event(1), 0);
 }
 
 }  // namespace
