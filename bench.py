@@ -115,7 +115,7 @@ def main():
                     "dtype": "bf16",
                     "data": "synthetic",
                     "config": {
-                        "model": f"GPT-1.3B ({n_params/1e9:.2f}B params, {args.model})",
+                        "model": f"GPT-{args.model} ({n_params/1e9:.2f}B params)",
                         "global_batch": args.batch * world * args.accum,
                         "seq_len": seq,
                         "parallelism": f"dp{world}+zero1",
