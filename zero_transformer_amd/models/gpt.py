@@ -70,12 +70,12 @@ class CausalSelfAttention(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         """Returns the PRE-dropout attention output; the residual dropout is
         fused into the Block's residual add (ops.residual_dropout_add)."""
-        qkv = F.linear(x, self.qkv_w)  # (B, T, 3C)
+        qkv = ops.linear(x, self.qkv_w)  # (B, T, 3C); wgrad on the side stream
         o = ops.attention_qkv(
             qkv, self.num_head, self.slopes,
             dropout_p=self.dropout_p, training=self.training,
         )
-        return self.fc_resid(o)
+        return ops.linear(o, self.fc_resid.weight)
 
     # .pth contract: expose query/key/value instead of the fused qkv_w
     def _save_to_state_dict(self, destination, prefix, keep_vars):
@@ -107,7 +107,7 @@ class MLP(nn.Module):
         self.fc_resid = nn.Linear(4 * dim, dim, bias=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.fc_resid(ops.gelu(self.fc1(x)))
+        return ops.linear(ops.gelu(ops.linear(x, self.fc1.weight)), self.fc_resid.weight)
 
 
 class Block(nn.Module):

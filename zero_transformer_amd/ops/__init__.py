@@ -52,6 +52,63 @@ def hip_available() -> bool:
 
 
 # ---------------------------------------------------------------------------
+# Linear with the weight gradient on a side stream
+# ---------------------------------------------------------------------------
+
+_WGRAD_STREAM: Optional["torch.cuda.Stream"] = None
+
+
+def wgrad_stream():
+    """Side stream for weight-gradient GEMMs (lazy, one per process).
+
+    Weight grads are off the critical path of backward (only the optimizer
+    consumes them): computing them here lets hipBLASLt fill the attention-
+    backward kernels' idle pipes. Ordering contract: every consumer of a
+    side-produced grad must wait_stream(wgrad_stream()) — the ZeRO
+    optimizer's adopt-hook does (parallel/zero.py _on_grad_ready).
+    """
+    global _WGRAD_STREAM
+    if _WGRAD_STREAM is None:
+        _WGRAD_STREAM = torch.cuda.Stream()
+    return _WGRAD_STREAM
+
+
+class _LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        return torch.nn.functional.linear(x, w)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dy @ w  # critical path, current stream
+        s = wgrad_stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            dw = dy.reshape(-1, dy.shape[-1]).T @ x.reshape(-1, x.shape[-1])
+        # keep x/dy alive for the side stream's GEMM
+        x.record_stream(s)
+        dy.record_stream(s)
+        # dw is handed to AccumulateGrad as a pointer assignment (p.grad is
+        # None between steps); the device-side consumer is the ZeRO hook's
+        # bucket copy, which waits on wgrad_stream first.
+        return dx, dw
+
+
+def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """F.linear whose weight gradient runs on the wgrad side stream (GPU).
+
+    Do NOT use for weight-tied layers (multiple grad contributions would
+    accumulate on the main stream unordered against the side stream).
+    """
+    if x.is_cuda and torch.is_grad_enabled() and w.requires_grad:
+        return _LinearFn.apply(x, w)
+    return torch.nn.functional.linear(x, w)
+
+
+# ---------------------------------------------------------------------------
 # LayerNorm (bias-free)
 # ---------------------------------------------------------------------------
 

@@ -155,6 +155,7 @@ class ZeRO1Optimizer:
 
     def _install_views_and_hooks(self):
         self._grad_view: Dict[int, torch.Tensor] = {}
+        self._accumulated: set = set()
         for b in self.buckets:
             for p, o in zip(b.params, b.offsets):
                 shape = p.data.shape
@@ -168,15 +169,26 @@ class ZeRO1Optimizer:
         self._sync = sync
 
     def _on_grad_ready(self, p: torch.nn.Parameter):
-        # First accumulation of the step: adopt autograd's freshly-assigned
-        # grad by copying it into the flat bucket view and re-pointing
-        # p.grad at the view, so later micro-steps accumulate in place.
-        # This replaces the old preset-view scheme, which paid a bucket-wide
-        # zero_() plus a read-modify-write add on the FIRST backward too.
+        # Adopt autograd's freshly-assigned grad into the flat bucket view
+        # (copy on first accumulation of the step, add on later micro-steps)
+        # and reset p.grad to None so every backward ASSIGNS a fresh tensor
+        # instead of read-modify-writing zeroed memory. Side-stream weight
+        # grads (ops.linear) are ordered here: wait wgrad_stream before the
+        # copy, and pin the grad's allocation to the current stream.
         view = self._grad_view[id(p)]
-        if p.grad is not view:
-            view.copy_(p.grad)
-            p.grad = view
+        g = p.grad
+        if g is not view:
+            if g.is_cuda:
+                from .. import ops as _ops
+
+                torch.cuda.current_stream(self.device).wait_stream(_ops.wgrad_stream())
+                g.record_stream(torch.cuda.current_stream(self.device))
+            if id(p) in self._accumulated:
+                view.add_(g)
+            else:
+                view.copy_(g)
+                self._accumulated.add(id(p))
+            p.grad = None
         if not self._sync:
             return
         b, _ = self._param_bucket[id(p)]
@@ -204,9 +216,8 @@ class ZeRO1Optimizer:
         for b in self.buckets:
             if b.ready != len(b.params):
                 for p in b.params:
-                    view = self._grad_view[id(p)]
-                    if p.grad is not view:
-                        view.zero_()
+                    if id(p) not in self._accumulated:
+                        self._grad_view[id(p)].zero_()
                 self._launch_reduce(b)
         if self.overlap_comm:
             torch.cuda.current_stream(self.device).wait_stream(self._comm_stream)
@@ -249,6 +260,7 @@ class ZeRO1Optimizer:
     def zero_grad(self, set_to_none: bool = False):
         # p.grad = None: the next backward ASSIGNS fresh grads (no add into
         # zeroed memory); _on_grad_ready copies them into the flat buckets.
+        self._accumulated.clear()
         for b in self.buckets:
             for p in b.params:
                 p.grad = None
