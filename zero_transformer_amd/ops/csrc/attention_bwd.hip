@@ -378,6 +378,301 @@ __global__ __launch_bounds__(512, 2) void flash_dkdv_kernel(
   }
 }
 
+// 4-wave variant of dQ (same rationale as flash_dkdv4_kernel: one wave per
+// SIMD with the whole register file — Q/dO row fragments resident, zero
+// spills). 128 q rows per block.
+template <int D>
+__global__ __launch_bounds__(256, 1) void flash_dq4_kernel(
+    const uint16_t* __restrict__ qkv, const uint16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    const float* __restrict__ slopes, uint16_t* __restrict__ dqkv, int H, int T,
+    int C, float scale, float p_drop, uint32_t seed) {
+  constexpr int RB4 = 128;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint16_t* k_lds = (uint16_t*)smem;
+  uint16_t* v_lds = k_lds + TB * 128;
+
+  const int bh = blockIdx.x;
+  const int h = bh % H;
+  const int QS = 3 * C;
+  const long base = (long)(bh / H) * T * QS + h * D;
+  const long kbase = base + C;
+  const long vbase = base + 2 * C;
+  const long dobase = (long)(bh / H) * T * C + h * D;
+  const int q0 = blockIdx.y * RB4;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int li = lane & 31, hi = lane >> 5;
+  const int qw = q0 + wave * 32;
+  const int qi = qw + li;
+  const float slope = slopes[h];
+  const uint32_t thr = (uint32_t)(p_drop * 256.0f + 0.5f);
+  const float inv_keep = thr ? 256.0f / (256.0f - (float)thr) : 1.0f;
+
+  constexpr int KS = D / 16;
+  constexpr int DB = D / 32;
+
+  bf16x8 q_frag[KS], do_frag[KS];
+  {
+    const bool ok = qi < T;
+#pragma unroll
+    for (int s = 0; s < KS; ++s) {
+      if (ok) {
+        q_frag[s] = *reinterpret_cast<const bf16x8*>(&qkv[base + (long)qi * QS + s * 16 + 8 * hi]);
+        do_frag[s] = *reinterpret_cast<const bf16x8*>(&dout[dobase + (long)qi * C + s * 16 + 8 * hi]);
+      } else {
+        q_frag[s] = bf16x8{};
+        do_frag[s] = bf16x8{};
+      }
+    }
+  }
+  const float lse_q = qi < T ? lse[(long)bh * T + qi] : 0.f;
+  const float delta_q = qi < T ? delta[(long)bh * T + qi] : 0.f;
+
+  f32x16 dq_acc[DB];
+#pragma unroll
+  for (int d = 0; d < DB; ++d) dq_acc[d] = f32x16{};
+
+  Stage<D, 256> sk, sv;
+  sk.load(qkv, kbase, QS, 0, T);
+  sv.load(qkv, vbase, QS, 0, T);
+  const int kv_end = min(T, q0 + RB4);
+  for (int kt = 0; kt < kv_end; kt += TB) {
+    __syncthreads();
+    sk.store(k_lds);
+    sv.store(v_lds);
+    __syncthreads();
+    attn::lds_acquire();
+    if (kt + TB < kv_end) {
+      sk.load(qkv, kbase, QS, kt + TB, T);
+      sv.load(qkv, vbase, QS, kt + TB, T);
+    }
+
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      const int kt32 = kt + sub * 32;
+      if (kt32 > qw + 31 || kt32 >= T) continue;
+
+      f32x16 s_acc{}, dpt_acc{};
+#pragma unroll
+      for (int s = 0; s < KS; ++s) {
+        const int kk = s * 16 + 8 * hi;
+        const int krow = sub * 32 + li;
+        bf16x8 ka = *reinterpret_cast<const bf16x8*>((char*)k_lds + swz(krow, krow * 256 + kk * 2));
+        bf16x8 va = *reinterpret_cast<const bf16x8*>((char*)v_lds + swz(krow, krow * 256 + kk * 2));
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, q_frag[s], s_acc, 0, 0, 0);
+        dpt_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, do_frag[s], dpt_acc, 0, 0, 0);
+      }
+
+      float ds[16];
+      if (thr) {
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+          const int kbase_g = kt32 + 8 * g + 4 * hi;
+          const uint32_t bits = drop_bits32(seed, bh * T + qi, kbase_g >> 2);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const int r = 4 * g + e;
+            const int kj = kbase_g + e;
+            const bool masked = kj > qi || kj >= T || qi >= T;
+            const float x = masked ? -3.0e38f
+                                   : s_acc[r] * scale + slope * (float)(kj - qi) - lse_q;
+            const float p = __expf(x);
+            const bool keep = ((bits >> (8 * e)) & 0xffu) >= thr;
+            const float dp = keep ? dpt_acc[r] * inv_keep : 0.f;
+            ds[r] = scale * p * (dp - delta_q);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kj = kt32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const bool masked = kj > qi || kj >= T || qi >= T;
+          const float x = masked ? -3.0e38f
+                                 : s_acc[r] * scale + slope * (float)(kj - qi) - lse_q;
+          ds[r] = scale * __expf(x) * (dpt_acc[r] - delta_q);
+        }
+      }
+
+      bf16x8 dsa[2];
+      c_to_a_frags(ds, dsa);
+
+#pragma unroll
+      for (int d = 0; d < DB; ++d) {
+        attn::TrPair kp = attn::tr_frag_pair(k_lds, sub * 32, d * 32);
+        dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[0], kp.a, dq_acc[d], 0, 0, 0);
+        dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[1], kp.b, dq_acc[d], 0, 0, 0);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const int qr = qw + row;
+    if (qr >= T) continue;
+#pragma unroll
+    for (int d = 0; d < DB; ++d)
+      dqkv[base + (long)qr * QS + d * 32 + li] = f32_to_bf16(dq_acc[d][r]);
+  }
+}
+
+// 4-wave variant of dKdV: 256 threads, ONE wave per SIMD, whole register
+// file per wave (no 2-waves/SIMD cap) — K/V resident AND no spills, at the
+// cost of losing the partner wave's latency hiding. A/B-selected against
+// the 8-wave kernel via ZTA_DKDV4=1 (see launch_bwd).
+template <int D>
+__global__ __launch_bounds__(256, 1) void flash_dkdv4_kernel(
+    const uint16_t* __restrict__ qkv, const uint16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    const float* __restrict__ slopes, uint16_t* __restrict__ dqkv, int H, int T,
+    int C, float scale, float p_drop, uint32_t seed) {
+  constexpr int RB4 = 128;  // 4 waves x 32 keys
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint16_t* q_lds = (uint16_t*)smem;
+  uint16_t* do_lds = q_lds + TB * 128;
+  float* lse_s = (float*)(do_lds + TB * 128);
+  float* delta_s = lse_s + TB;
+
+  const int bh = blockIdx.x;
+  const int h = bh % H;
+  const int QS = 3 * C;
+  const long base = (long)(bh / H) * T * QS + h * D;
+  const long kbase = base + C;
+  const long vbase = base + 2 * C;
+  const long dobase = (long)(bh / H) * T * C + h * D;
+  const int k0 = blockIdx.y * RB4;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int li = lane & 31, hi = lane >> 5;
+  const int kw = k0 + wave * 32;
+  const int kj = kw + li;
+  const float slope = slopes[h];
+  const uint32_t thr = (uint32_t)(p_drop * 256.0f + 0.5f);
+  const float inv_keep = thr ? 256.0f / (256.0f - (float)thr) : 1.0f;
+
+  constexpr int KS = D / 16;
+  constexpr int DB = D / 32;
+
+  bf16x8 k_frag[KS], v_frag[KS];
+  {
+    const bool ok = kj < T;
+#pragma unroll
+    for (int s = 0; s < KS; ++s) {
+      if (ok) {
+        k_frag[s] = *reinterpret_cast<const bf16x8*>(&qkv[kbase + (long)kj * QS + s * 16 + 8 * hi]);
+        v_frag[s] = *reinterpret_cast<const bf16x8*>(&qkv[vbase + (long)kj * QS + s * 16 + 8 * hi]);
+      } else {
+        k_frag[s] = bf16x8{};
+        v_frag[s] = bf16x8{};
+      }
+    }
+  }
+
+  f32x16 dk_acc[DB], dv_acc[DB];
+#pragma unroll
+  for (int d = 0; d < DB; ++d) {
+    dk_acc[d] = f32x16{};
+    dv_acc[d] = f32x16{};
+  }
+
+  Stage<D, 256> sq, sdo;
+  float lse_reg = 0.f, delta_reg = 0.f;
+  const int t256 = threadIdx.x;
+  auto load_stats = [&](int qt) {
+    if (t256 < TB) {
+      const int qg = qt + t256;
+      lse_reg = qg < T ? lse[(long)bh * T + qg] : 0.f;
+      delta_reg = qg < T ? delta[(long)bh * T + qg] : 0.f;
+    }
+  };
+
+  sq.load(qkv, base, QS, k0, T);
+  sdo.load(dout, dobase, C, k0, T);
+  load_stats(k0);
+  for (int qt = k0; qt < T; qt += TB) {
+    __syncthreads();
+    sq.store(q_lds);
+    sdo.store(do_lds);
+    if (t256 < TB) {
+      lse_s[t256] = lse_reg;
+      delta_s[t256] = delta_reg;
+    }
+    __syncthreads();
+    attn::lds_acquire();
+    if (qt + TB < T) {
+      sq.load(qkv, base, QS, qt + TB, T);
+      sdo.load(dout, dobase, C, qt + TB, T);
+      load_stats(qt + TB);
+    }
+
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      const int q32 = qt + sub * 32;
+      if (q32 + 31 < kw || q32 >= T) continue;
+
+      f32x16 s_acc{}, dp_acc{};
+#pragma unroll
+      for (int s = 0; s < KS; ++s) {
+        const int kk = s * 16 + 8 * hi;
+        const int qrow = sub * 32 + li;
+        bf16x8 qa = *reinterpret_cast<const bf16x8*>((char*)q_lds + swz(qrow, qrow * 256 + kk * 2));
+        bf16x8 da = *reinterpret_cast<const bf16x8*>((char*)do_lds + swz(qrow, qrow * 256 + kk * 2));
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, k_frag[s], s_acc, 0, 0, 0);
+        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, v_frag[s], dp_acc, 0, 0, 0);
+      }
+
+      float p_pv[16], ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const int qi = q32 + row;
+        const float lq = lse_s[sub * 32 + row];
+        const float dq_ = delta_s[sub * 32 + row];
+        const bool masked = kj > qi || kj >= T || qi >= T;
+        const float x = masked ? -3.0e38f
+                               : s_acc[r] * scale + slope * (float)(kj - qi) - lq;
+        const float p = __expf(x);
+        float dp = dp_acc[r];
+        float ppv = p;
+        if (thr) {
+          const uint32_t bits = drop_bits32(seed, bh * T + qi, kj >> 2);
+          const bool keep = ((bits >> (8 * (kj & 3))) & 0xffu) >= thr;
+          dp = keep ? dp * inv_keep : 0.f;
+          ppv = keep ? p * inv_keep : 0.f;
+        }
+        p_pv[r] = ppv;
+        ds[r] = scale * p * (dp - dq_);
+      }
+
+      bf16x8 pa[2], dsa[2];
+      c_to_a_frags(p_pv, pa);
+      c_to_a_frags(ds, dsa);
+
+#pragma unroll
+      for (int d = 0; d < DB; ++d) {
+        attn::TrQuad t = attn::tr_frag_quad(do_lds, q_lds, sub * 32, d * 32);
+        dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], t.x.a, dv_acc[d], 0, 0, 0);
+        dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], t.x.b, dv_acc[d], 0, 0, 0);
+        dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[0], t.y.a, dk_acc[d], 0, 0, 0);
+        dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[1], t.y.b, dk_acc[d], 0, 0, 0);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const int kr = kw + row;
+    if (kr >= T) continue;
+#pragma unroll
+    for (int d = 0; d < DB; ++d) {
+      dqkv[kbase + (long)kr * QS + d * 32 + li] = f32_to_bf16(dk_acc[d][r]);
+      dqkv[vbase + (long)kr * QS + d * 32 + li] = f32_to_bf16(dv_acc[d][r]);
+    }
+  }
+}
+
 // dq and dkdv are independent (disjoint column ranges of dqkv) and both
 // stall-bound at one 512-thread block per CU — run them CONCURRENTLY on two
 // streams so their blocks co-reside (227 + 256 VGPR and 32 + 64.5 KB LDS
@@ -414,12 +709,37 @@ void launch_bwd(const at::Tensor& qkv, const at::Tensor& dout, const at::Tensor&
   (void)hipEventRecord(bwd_event(0), stream);
   (void)hipStreamWaitEvent(side, bwd_event(0), 0);
   const size_t smem_dq = 2 * TB * 128 * sizeof(uint16_t);
+  static const bool dq4 = [] {
+    const char* e = getenv("ZTA_DQ4");
+    return e && e[0] == '1';  // default OFF: measured slower than the 8-wave
+                              // dq when concurrent with the 4-wave dKdV
+  }();
+  if (dq4) {
+    dim3 gridq(B * H, (T + 127) / 128);
+    hipLaunchKernelGGL(flash_dq4_kernel<D>, gridq, dim3(256), smem_dq, stream,
+                       (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,
+                       scale, p_drop, seed);
+  } else
   hipLaunchKernelGGL(flash_dq_kernel<D>, grid, dim3(512), smem_dq, stream,
                      (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
                      slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,
                      scale, p_drop, seed);
   const size_t smem_kv = 2 * TB * 128 * sizeof(uint16_t) + 2 * TB * sizeof(float);
+  static const bool use4 = [] {
+    const char* e = getenv("ZTA_DKDV4");
+    return !(e && e[0] == '0');  // default ON (zero spills beat 2 waves/SIMD)
+  }();
+  if (use4) {
+    dim3 grid4(B * H, (T + 127) / 128);
+    hipLaunchKernelGGL(flash_dkdv4_kernel<D>, grid4, dim3(256), smem_kv, side,
+                       (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,
+                       scale, p_drop, seed);
+  } else
   hipLaunchKernelGGL(flash_dkdv_kernel<D>, grid, dim3(512), smem_kv, side,
                      (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),

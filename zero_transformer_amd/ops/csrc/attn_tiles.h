@@ -163,9 +163,9 @@ ZTA_DEV void c_to_a_frags(const float* x, bf16x8* pa) {
 // swizzled LDS image: issue global loads into registers early (hide HBM
 // latency under the previous tile's compute), write to LDS after the
 // barrier. Thread t owns elements {t*8 + c*4096 | c}, row = idx/D.
-template <int D>
+template <int D, int NT = 512>
 struct Stage {
-  static constexpr int NC = (TB * D + 512 * 8 - 1) / (512 * 8);
+  static constexpr int NC = (TB * D + NT * 8 - 1) / (NT * 8);
   s16x8 r[NC];
   // `base` points at row 0 of this (b, h) plane; `rs` is the row stride in
   // elements (3*C for tensors packed as (B, T, 3C) qkv, C for (B, T, C)).
@@ -173,7 +173,7 @@ struct Stage {
     const int t = threadIdx.x;
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
-      const int idx = t * 8 + c * 512 * 8;
+      const int idx = t * 8 + c * NT * 8;
       const int rg = row0 + idx / D;
       r[c] = (idx < TB * D && rg < T)
                  ? *reinterpret_cast<const s16x8*>(&g[base + (long)rg * rs + idx % D])
@@ -184,7 +184,7 @@ struct Stage {
     const int t = threadIdx.x;
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
-      const int idx = t * 8 + c * 512 * 8;
+      const int idx = t * 8 + c * NT * 8;
       if (idx >= TB * D) break;
       const int row = idx / D, d = idx % D;
       *reinterpret_cast<s16x8*>((char*)lds + swz(row, row * 256 + d * 2)) = r[c];
