@@ -23,10 +23,14 @@ def main():
     p.add_argument("--models", nargs="+", default=["1_3b_2048"])
     p.add_argument("--batch", type=int, default=8)
     p.add_argument("--steps", type=int, default=2)
+    p.add_argument("--duration", type=int, default=100, help="ms per candidate algo")
+    p.add_argument("--iters", type=int, default=0, help="max tuning iterations (0: default)")
     args = p.parse_args()
     assert torch.cuda.is_available()
     assert gemm_tune.enable(tuning=True), "TunableOp unavailable"
-    torch.cuda.tunable.set_max_tuning_duration(100)
+    torch.cuda.tunable.set_max_tuning_duration(args.duration)
+    if args.iters:
+        torch.cuda.tunable.set_max_tuning_iterations(args.iters)
 
     from zero_transformer_amd.models import model_getter
     from zero_transformer_amd.parallel.zero import ZeRO1Optimizer
