@@ -323,3 +323,20 @@ def test_model_train_step_gpu(dev):
     losses = [eng.train_step(batch)["train/loss"] for _ in range(5)]
     assert all(math.isfinite(l) for l in losses)
     assert losses[-1] < losses[0], f"loss did not decrease: {losses}"
+
+
+def test_inference_kv_cache_gpu(dev):
+    """KV-cached inference model on GPU: cached decode == full forward."""
+    from zero_transformer_amd.models.inference import GPT2
+
+    torch.manual_seed(12)
+    model = GPT2(embedding_dim=256, vocab_size=512, num_head=4, num_ctx=64, N=2).to(dev).eval()
+    idx = torch.randint(0, 512, (2, 16), device=dev)
+    with torch.no_grad():
+        full = model(idx)
+        # prefill then decode one token with the cache
+        logits, presents = model(idx[:, :-1], use_cache=True)
+        step, _ = model(idx[:, -1:], use_cache=True, past_states=presents)
+    assert torch.allclose(step[:, -1], full[:, -1], atol=1e-3, rtol=1e-3)
+    out = model.generate(idx, max_new_tokens=8)
+    assert out.shape == (2, 24)

@@ -48,8 +48,8 @@ constexpr float NEG_INF = -3.0e38f;
 // offset h*D, k at C + h*D, v at 2C + h*D); o is (B, T, C). This is the
 // natural layout of the model's single qkv GEMM — no transposes or
 // .contiguous() copies on either side of the op.
-template <int D>
-__global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
+template <int D, int NT, int MINW>
+__global__ __launch_bounds__(NT, MINW) void flash_fwd_kernel(
     const uint16_t* __restrict__ qkv, const float* __restrict__ slopes,
     uint16_t* __restrict__ o, float* __restrict__ lse, int H, int T, int C,
     float scale, float p_drop, uint32_t seed) {
@@ -65,7 +65,8 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
   const long kbase = base + C;
   const long vbase = base + 2 * C;
   const long obase = (long)(bh / H) * T * C + h * D;   // o plane
-  const int q0 = blockIdx.y * RB;
+  constexpr int RBX = NT / 2;  // NT=512: 256 q rows; NT=256: 128 q rows
+  const int q0 = blockIdx.y * RBX;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int li = lane & 31;
@@ -96,13 +97,13 @@ __global__ __launch_bounds__(512, 2) void flash_fwd_kernel(
 #pragma unroll
   for (int d = 0; d < DB; ++d) o_acc[d] = f32x16{};
 
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+  if (NT == 512 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);  // static priority for the younger half (T5)
 
-  Stage<D> sk, sv;
+  Stage<D, NT> sk, sv;
   sk.load(qkv, kbase, QS, 0, T);
   sv.load(qkv, vbase, QS, 0, T);
-  const int kv_end = min(T, q0 + RB);
+  const int kv_end = min(T, q0 + RBX);
   for (int kt = 0; kt < kv_end; kt += TB) {
     __syncthreads();  // previous tile fully consumed
     sk.store(k_lds);
@@ -220,9 +221,21 @@ template <int D>
 void launch_fwd(const at::Tensor& qkv, const at::Tensor& slopes, at::Tensor& o,
                 at::Tensor& lse, int B, int H, int T, int C, float scale,
                 float p_drop, uint32_t seed, hipStream_t stream) {
-  dim3 grid(B * H, (T + RB - 1) / RB);
   const size_t smem = 2 * TB * 128 * sizeof(uint16_t);
-  hipLaunchKernelGGL(flash_fwd_kernel<D>, grid, dim3(512), smem, stream,
+  static const bool f4 = [] {
+    const char* e = getenv("ZTA_FWD4");
+    return e && e[0] == '1';
+  }();
+  if (f4) {
+    dim3 grid(B * H, (T + 127) / 128);
+    hipLaunchKernelGGL((flash_fwd_kernel<D, 256, 1>), grid, dim3(256), smem, stream,
+                       (const uint16_t*)qkv.data_ptr(), slopes.data_ptr<float>(),
+                       (uint16_t*)o.data_ptr(), lse.data_ptr<float>(), H, T, C,
+                       scale, p_drop, seed);
+    return;
+  }
+  dim3 grid(B * H, (T + RB - 1) / RB);
+  hipLaunchKernelGGL((flash_fwd_kernel<D, 512, 2>), grid, dim3(512), smem, stream,
                      (const uint16_t*)qkv.data_ptr(), slopes.data_ptr<float>(),
                      (uint16_t*)o.data_ptr(), lse.data_ptr<float>(), H, T, C,
                      scale, p_drop, seed);
