@@ -78,8 +78,8 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
     const float* __restrict__ slopes, uint16_t* __restrict__ dqkv, int H, int T,
     int C, float scale, float p_drop, uint32_t seed) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  uint16_t* k_lds = (uint16_t*)smem;
-  uint16_t* v_lds = k_lds + TB * 128;
+  // double-buffered LDS-DMA staging: layout [k0 | v0 | k1 | v1]
+  uint16_t* lds0 = (uint16_t*)smem;
 
   const int bh = blockIdx.x;  // grid: (BH, tiles) for per-CU load balance
   const int h = bh % H;
@@ -125,19 +125,22 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);  // static priority for the younger half (T5)
 
-  Stage<D> sk, sv;
-  sk.load(qkv, kbase, QS, 0, T);
-  sv.load(qkv, vbase, QS, 0, T);
+  attn::glds_stage<D, 8>(qkv, kbase, QS, 0, T, lds0);
+  attn::glds_stage<D, 8>(qkv, vbase, QS, 0, T, lds0 + TB * 128);
   const int kv_end = min(T, q0 + RB);
   for (int kt = 0; kt < kv_end; kt += TB) {
-    __syncthreads();  // previous tile fully consumed
-    sk.store(k_lds);
-    sv.store(v_lds);
-    __syncthreads();  // tile kt visible
+    const int cur = (kt / TB) & 1;
+    const uint16_t* k_lds = lds0 + (cur ? 2 * TB * 128 : 0);
+    const uint16_t* v_lds = lds0 + TB * 128 + (cur ? 2 * TB * 128 : 0);
+    // ONE barrier per tile: waits this tile's in-flight LDS-DMA (vmcnt)
+    // AND guarantees everyone is done reading the buffer the next
+    // prefetch overwrites (last read two iterations ago).
+    __syncthreads();
     attn::lds_acquire();
     if (kt + TB < kv_end) {
-      sk.load(qkv, kbase, QS, kt + TB, T);
-      sv.load(qkv, vbase, QS, kt + TB, T);
+      uint16_t* nxt = lds0 + (cur ? 0 : 2 * TB * 128);
+      attn::glds_stage<D, 8>(qkv, kbase, QS, kt + TB, T, nxt);
+      attn::glds_stage<D, 8>(qkv, vbase, QS, kt + TB, T, nxt + TB * 128);
     }
 
 #pragma unroll
@@ -708,7 +711,8 @@ void launch_bwd(const at::Tensor& qkv, const at::Tensor& dout, const at::Tensor&
   hipStream_t side = bwd_side_stream();
   (void)hipEventRecord(bwd_event(0), stream);
   (void)hipStreamWaitEvent(side, bwd_event(0), 0);
-  const size_t smem_dq = 2 * TB * 128 * sizeof(uint16_t);
+  const size_t smem_dq = 4 * TB * 128 * sizeof(uint16_t);  // 2 tensors x 2 buffers
+  const size_t smem_dq4 = 2 * TB * 128 * sizeof(uint16_t);
   static const bool dq4 = [] {
     const char* e = getenv("ZTA_DQ4");
     return e && e[0] == '1';  // default OFF: measured slower than the 8-wave
@@ -716,7 +720,7 @@ void launch_bwd(const at::Tensor& qkv, const at::Tensor& dout, const at::Tensor&
   }();
   if (dq4) {
     dim3 gridq(B * H, (T + 127) / 128);
-    hipLaunchKernelGGL(flash_dq4_kernel<D>, gridq, dim3(256), smem_dq, stream,
+    hipLaunchKernelGGL(flash_dq4_kernel<D>, gridq, dim3(256), smem_dq4, stream,
                        (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,

@@ -159,6 +159,36 @@ ZTA_DEV void c_to_a_frags(const float* x, bf16x8* pa) {
   pa[1] = cvt.v8;
 }
 
+// Direct LDS-DMA staging (global_load_lds_dwordx4) of a 64-row x D-col bf16
+// tile into the 256 B-stride XOR-swizzled image: no staging registers, no
+// ds_write phase, and (with double-buffered LDS) ONE barrier per tile.
+// Semantics measured by tools/probes/glds_probe.hip: the LDS base is
+// wave-uniform and lane l's 16 bytes land at base + 16*l, so one
+// wave-instruction fills 1 KiB = four 256 B image rows; the per-lane SOURCE
+// address carries the swizzle (dest slot s of row r holds source columns
+// 8*(s ^ (r&7))..+7 — T2 note: with lane-linear destinations the XOR moves
+// to the source side). Rows beyond T are clamped to row T-1: the data is
+// finite real-tensor content and every consumer masks those rows to p = 0.
+// NWAVES = threads/64; chunk k (4 rows) is staged by wave k % NWAVES.
+template <int D, int NWAVES>
+ZTA_DEV void glds_stage(const uint16_t* g, long base, int rs, int row0, int T,
+                        uint16_t* lds) {
+  const int wave = (threadIdx.x >> 6);
+  const int l = threadIdx.x & 63;
+  const int s = l & 15;        // 16 B slot within the 256 B image row
+  const int rsub = l >> 4;     // row within the chunk's 4 rows
+#pragma unroll
+  for (int k = wave; k < 16; k += NWAVES) {
+    const int row = 4 * k + rsub;
+    int c8 = 8 * (s ^ (row & 7));  // swizzled source column block
+    if (D < 128 && c8 >= D) c8 = 0;  // slot never read for cols >= D
+    const int rg = min(row0 + row, T - 1);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)&g[base + (long)rg * rs + c8],
+        (__attribute__((address_space(3))) void*)(lds + (long)k * 512), 16, 0, 0);
+  }
+}
+
 // T14 tile staging of a 64-row x D-col bf16 tile into a 256 B-stride
 // swizzled LDS image: issue global loads into registers early (hide HBM
 // latency under the previous tile's compute), write to LDS after the
