@@ -532,10 +532,9 @@ __global__ __launch_bounds__(256, 1) void flash_dkdv4_kernel(
     int C, float scale, float p_drop, uint32_t seed) {
   constexpr int RB4 = 128;  // 4 waves x 32 keys
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  uint16_t* q_lds = (uint16_t*)smem;
-  uint16_t* do_lds = q_lds + TB * 128;
-  float* lse_s = (float*)(do_lds + TB * 128);
-  float* delta_s = lse_s + TB;
+  // LDS-DMA double buffers [q0 | do0 | q1 | do1] + 2x (lse, delta) slices
+  uint16_t* lds0 = (uint16_t*)smem;
+  float* stats0 = (float*)(lds0 + 4 * TB * 128);
 
   const int bh = blockIdx.x;
   const int h = bh % H;
@@ -579,34 +578,32 @@ __global__ __launch_bounds__(256, 1) void flash_dkdv4_kernel(
     dv_acc[d] = f32x16{};
   }
 
-  Stage<D, 256> sq, sdo;
-  float lse_reg = 0.f, delta_reg = 0.f;
   const int t256 = threadIdx.x;
-  auto load_stats = [&](int qt) {
+  auto stage_stats = [&](int qt, float* dst) {
+    // lse slice then delta slice, plain LDS stores (tiny)
     if (t256 < TB) {
       const int qg = qt + t256;
-      lse_reg = qg < T ? lse[(long)bh * T + qg] : 0.f;
-      delta_reg = qg < T ? delta[(long)bh * T + qg] : 0.f;
+      dst[t256] = qg < T ? lse[(long)bh * T + qg] : 0.f;
+      dst[TB + t256] = qg < T ? delta[(long)bh * T + qg] : 0.f;
     }
   };
 
-  sq.load(qkv, base, QS, k0, T);
-  sdo.load(dout, dobase, C, k0, T);
-  load_stats(k0);
+  attn::glds_stage<D, 4>(qkv, base, QS, k0, T, lds0);
+  attn::glds_stage<D, 4>(dout, dobase, C, k0, T, lds0 + TB * 128);
+  stage_stats(k0, stats0);
   for (int qt = k0; qt < T; qt += TB) {
-    __syncthreads();
-    sq.store(q_lds);
-    sdo.store(do_lds);
-    if (t256 < TB) {
-      lse_s[t256] = lse_reg;
-      delta_s[t256] = delta_reg;
-    }
+    const int cur = ((qt - k0) / TB) & 1;
+    const uint16_t* q_lds = lds0 + (cur ? 2 * TB * 128 : 0);
+    const uint16_t* do_lds = lds0 + TB * 128 + (cur ? 2 * TB * 128 : 0);
+    const float* lse_s = stats0 + (cur ? 2 * TB : 0);
+    const float* delta_s = lse_s + TB;
     __syncthreads();
     attn::lds_acquire();
     if (qt + TB < T) {
-      sq.load(qkv, base, QS, qt + TB, T);
-      sdo.load(dout, dobase, C, qt + TB, T);
-      load_stats(qt + TB);
+      uint16_t* nxt = lds0 + (cur ? 0 : 2 * TB * 128);
+      attn::glds_stage<D, 4>(qkv, base, QS, qt + TB, T, nxt);
+      attn::glds_stage<D, 4>(dout, dobase, C, qt + TB, T, nxt + TB * 128);
+      stage_stats(qt + TB, stats0 + (cur ? 0 : 2 * TB));
     }
 
 #pragma unroll
@@ -732,13 +729,14 @@ void launch_bwd(const at::Tensor& qkv, const at::Tensor& dout, const at::Tensor&
                      slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,
                      scale, p_drop, seed);
   const size_t smem_kv = 2 * TB * 128 * sizeof(uint16_t) + 2 * TB * sizeof(float);
+  const size_t smem_kv4 = 4 * TB * 128 * sizeof(uint16_t) + 4 * TB * sizeof(float);
   static const bool use4 = [] {
     const char* e = getenv("ZTA_DKDV4");
     return !(e && e[0] == '0');  // default ON (zero spills beat 2 waves/SIMD)
   }();
   if (use4) {
     dim3 grid4(B * H, (T + 127) / 128);
-    hipLaunchKernelGGL(flash_dkdv4_kernel<D>, grid4, dim3(256), smem_kv, side,
+    hipLaunchKernelGGL(flash_dkdv4_kernel<D>, grid4, dim3(256), smem_kv4, side,
                        (const uint16_t*)qkv.data_ptr(), (const uint16_t*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        slopes.data_ptr<float>(), (uint16_t*)dqkv.data_ptr(), H, T, C,

@@ -54,8 +54,8 @@ __global__ __launch_bounds__(NT, MINW) void flash_fwd_kernel(
     uint16_t* __restrict__ o, float* __restrict__ lse, int H, int T, int C,
     float scale, float p_drop, uint32_t seed) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  uint16_t* k_lds = (uint16_t*)smem;       // 64 x [256 B], swizzled
-  uint16_t* v_lds = k_lds + TB * 128;      // 64 x [256 B], swizzled
+  // double-buffered LDS-DMA staging: layout [k0 | v0 | k1 | v1]
+  uint16_t* lds0 = (uint16_t*)smem;
 
   const int bh = blockIdx.x;  // grid: (BH, tiles) — consecutive blocks share the
   // tile index so per-CU work is balanced across the causal triangle
@@ -100,19 +100,21 @@ __global__ __launch_bounds__(NT, MINW) void flash_fwd_kernel(
   if (NT == 512 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);  // static priority for the younger half (T5)
 
-  Stage<D, NT> sk, sv;
-  sk.load(qkv, kbase, QS, 0, T);
-  sv.load(qkv, vbase, QS, 0, T);
+  attn::glds_stage<D, NT / 64>(qkv, kbase, QS, 0, T, lds0);
+  attn::glds_stage<D, NT / 64>(qkv, vbase, QS, 0, T, lds0 + TB * 128);
   const int kv_end = min(T, q0 + RBX);
   for (int kt = 0; kt < kv_end; kt += TB) {
-    __syncthreads();  // previous tile fully consumed
-    sk.store(k_lds);
-    sv.store(v_lds);
-    __syncthreads();  // tile kt visible
+    const int cur = (kt / TB) & 1;
+    const uint16_t* k_lds = lds0 + (cur ? 2 * TB * 128 : 0);
+    const uint16_t* v_lds = lds0 + TB * 128 + (cur ? 2 * TB * 128 : 0);
+    // one barrier per tile: waits in-flight LDS-DMA and protects the
+    // buffer the next prefetch overwrites
+    __syncthreads();
     attn::lds_acquire();
     if (kt + TB < kv_end) {
-      sk.load(qkv, kbase, QS, kt + TB, T);
-      sv.load(qkv, vbase, QS, kt + TB, T);
+      uint16_t* nxt = lds0 + (cur ? 0 : 2 * TB * 128);
+      attn::glds_stage<D, NT / 64>(qkv, kbase, QS, kt + TB, T, nxt);
+      attn::glds_stage<D, NT / 64>(qkv, vbase, QS, kt + TB, T, nxt + TB * 128);
     }
 
 #pragma unroll
@@ -221,7 +223,7 @@ template <int D>
 void launch_fwd(const at::Tensor& qkv, const at::Tensor& slopes, at::Tensor& o,
                 at::Tensor& lse, int B, int H, int T, int C, float scale,
                 float p_drop, uint32_t seed, hipStream_t stream) {
-  const size_t smem = 2 * TB * 128 * sizeof(uint16_t);
+  const size_t smem = 4 * TB * 128 * sizeof(uint16_t);  // 2 tensors x 2 buffers
   static const bool f4 = [] {
     const char* e = getenv("ZTA_FWD4");
     return e && e[0] == '1';
