@@ -649,9 +649,19 @@ __global__ __launch_bounds__(256, 1) void flash_dkdv4_kernel(
       c_to_a_frags(p_pv, pa);
       c_to_a_frags(ds, dsa);
 
+      // software-pipelined transpose reads: issue d+1's eight reads before
+      // d's MFMAs, wait counted (the newer group's 8 ops may stay in flight)
+      attn::TrQuad qq[2];
+      attn::tr_quad_issue(do_lds, q_lds, sub * 32, 0, &qq[0]);
 #pragma unroll
       for (int d = 0; d < DB; ++d) {
-        attn::TrQuad t = attn::tr_frag_quad(do_lds, q_lds, sub * 32, d * 32);
+        if (d + 1 < DB)
+          attn::tr_quad_issue(do_lds, q_lds, sub * 32, (d + 1) * 32, &qq[(d + 1) & 1]);
+        attn::TrQuad& t = qq[d & 1];
+        if (d + 1 < DB)
+          attn::tr_quad_wait<8>(&t);
+        else
+          attn::tr_quad_wait<0>(&t);
         dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], t.x.a, dv_acc[d], 0, 0, 0);
         dv_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], t.x.b, dv_acc[d], 0, 0, 0);
         dk_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[0], t.y.a, dk_acc[d], 0, 0, 0);

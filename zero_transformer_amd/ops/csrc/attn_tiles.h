@@ -106,6 +106,57 @@ ZTA_DEV TrQuad tr_frag_quad(const uint16_t* lds_x, const uint16_t* lds_y, int k0
         "v"(ay[2]), "v"(ay[3]));
   return u.f;
 }
+// Software-pipelined form of tr_frag_quad: tr_quad_issue launches the eight
+// reads with NO wait; tr_quad_wait waits `lgkmcnt(N)` with the fragment
+// registers as pass-through operands so every consumer orders after it.
+// Safety: DS ops complete IN ORDER, so waiting with N = (number of DS ops
+// issued after this group) is correct even if the compiler interleaves its
+// own LDS ops — any overcount only waits longer. (SMEM also counts in
+// lgkmcnt and completes out of order, but none is live in these loops.)
+ZTA_DEV void tr_quad_issue(const uint16_t* lds_x, const uint16_t* lds_y, int k0,
+                           int j0, TrQuad* out) {
+  const int l = threadIdx.x & 63;
+  const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
+  const int rb = 8 * (l >> 5) + ((l >> 2) & 3);
+  int ax[4], ay[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int row = k0 + 16 * (i >> 1) + 4 * (i & 1) + rb;
+    const int off = row * 256 + (colb ^ ((row & 7) << 4));
+    ax[i] = (int)(size_t)((const char*)lds_x + off);
+    ay[i] = (int)(size_t)((const char*)lds_y + off);
+  }
+  union U {
+    i32x2 d[8];
+    TrQuad f;
+  }* u = (union U*)out;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %8\n\t"
+      "ds_read_b64_tr_b16 %1, %9\n\t"
+      "ds_read_b64_tr_b16 %2, %10\n\t"
+      "ds_read_b64_tr_b16 %3, %11\n\t"
+      "ds_read_b64_tr_b16 %4, %12\n\t"
+      "ds_read_b64_tr_b16 %5, %13\n\t"
+      "ds_read_b64_tr_b16 %6, %14\n\t"
+      "ds_read_b64_tr_b16 %7, %15"
+      : "=&v"(u->d[0]), "=&v"(u->d[1]), "=&v"(u->d[2]), "=&v"(u->d[3]),
+        "=&v"(u->d[4]), "=&v"(u->d[5]), "=&v"(u->d[6]), "=&v"(u->d[7])
+      : "v"(ax[0]), "v"(ax[1]), "v"(ax[2]), "v"(ax[3]), "v"(ay[0]), "v"(ay[1]),
+        "v"(ay[2]), "v"(ay[3]));
+}
+
+template <int N>
+ZTA_DEV void tr_quad_wait(TrQuad* q) {
+  union U {
+    i32x2 d[8];
+    TrQuad f;
+  }* u = (union U*)q;
+  asm volatile("s_waitcnt lgkmcnt(%8)"
+               : "+v"(u->d[0]), "+v"(u->d[1]), "+v"(u->d[2]), "+v"(u->d[3]),
+                 "+v"(u->d[4]), "+v"(u->d[5]), "+v"(u->d[6]), "+v"(u->d[7])
+               : "i"(N));
+}
+
 ZTA_DEV TrPair tr_frag_pair(const uint16_t* lds, int k0, int j0) {
   const int l = threadIdx.x & 63;
   const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
