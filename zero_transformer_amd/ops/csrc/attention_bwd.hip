@@ -19,9 +19,13 @@
 //       in-register and Q^T / dO^T B-fragments via tr-reads.
 //
 // The dropout mask M regenerates from the counter RNG (common.h
-// drop_bits32) with the forward's seed. All softmax math fp32. Staging is
-// T14 async-split: next tile's global loads issue before the compute phase,
-// LDS writes land after the barrier.
+// drop_bits32) with the forward's seed. All softmax math fp32. Tile staging
+// is direct LDS-DMA (global_load_lds_dwordx4, attn::glds_stage) into
+// double-buffered swizzled images: the prefetch for tile t+1 issues right
+// after tile t becomes visible, one barrier per tile. The default dKdV is
+// the 4-wave whole-register-file variant (flash_dkdv4_kernel, zero spills)
+// running concurrently with dQ on a side stream; the 8-wave kernels remain
+// selectable via ZTA_DKDV4=0 / ZTA_DQ4=1 for A/B.
 
 #include "attn_tiles.h"
 #include "common.h"

@@ -10,8 +10,10 @@
 //
 // Structure (the CDNA4 guide's swapped-QK^T recipe, §B attention):
 //   * 8 waves/block, each owning 32 q rows (QB = 256); K/V staged per
-//     64-key tile, row-major XOR-swizzled (T2), with T14 async staging:
-//     next tile's global loads issue before this tile's compute.
+//     64-key tile into double-buffered row-major XOR-swizzled (T2) LDS
+//     images by direct LDS-DMA (global_load_lds_dwordx4, attn::glds_stage):
+//     the next tile's DMA issues while this tile computes, one barrier per
+//     tile and no staging registers.
 //   * QK^T is computed SWAPPED — mfma(A=K, B=Q^T) — so the C column index
 //     is the q row: each lane holds a full P-row segment in registers and
 //     the softmax row-reduce is in-lane + one shfl_xor(32), with scalar
