@@ -197,12 +197,19 @@ __global__ __launch_bounds__(512, 2) void flash_dq_kernel(
       bf16x8 dsa[2];
       c_to_a_frags(ds, dsa);  // A[i=q][k=key]
 
-      // dQ += dS K : B[k=key][j=d] via tr-read of the row-major K tile
+      // dQ += dS K : B[k=key][j=d] via pipelined tr-reads of the K tile
+      attn::TrPair kp[2];
+      attn::tr_pair_issue(k_lds, sub * 32, 0, &kp[0]);
 #pragma unroll
       for (int d = 0; d < DB; ++d) {
-        attn::TrPair kp = attn::tr_frag_pair(k_lds, sub * 32, d * 32);
-        dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[0], kp.a, dq_acc[d], 0, 0, 0);
-        dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[1], kp.b, dq_acc[d], 0, 0, 0);
+        if (d + 1 < DB) attn::tr_pair_issue(k_lds, sub * 32, (d + 1) * 32, &kp[(d + 1) & 1]);
+        attn::TrPair& t = kp[d & 1];
+        if (d + 1 < DB)
+          attn::tr_pair_wait<4>(&t);
+        else
+          attn::tr_pair_wait<0>(&t);
+        dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[0], t.a, dq_acc[d], 0, 0, 0);
+        dq_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[1], t.b, dq_acc[d], 0, 0, 0);
       }
     }
   }

@@ -195,12 +195,19 @@ __global__ __launch_bounds__(NT, MINW) void flash_fwd_kernel(
       bf16x8 pa[2];
       attn::c_to_a_frags(p, pa);
 
-      // ---- O += P @ V : V^T B-fragments via tr-read of row-major V ----
+      // ---- O += P @ V : V^T B-fragments via pipelined tr-reads ----
+      attn::TrPair vp[2];
+      attn::tr_pair_issue(v_lds, sub * 32, 0, &vp[0]);
 #pragma unroll
       for (int d = 0; d < DB; ++d) {
-        attn::TrPair vp = attn::tr_frag_pair(v_lds, sub * 32, d * 32);
-        o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], vp.a, o_acc[d], 0, 0, 0);
-        o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], vp.b, o_acc[d], 0, 0, 0);
+        if (d + 1 < DB) attn::tr_pair_issue(v_lds, sub * 32, (d + 1) * 32, &vp[(d + 1) & 1]);
+        attn::TrPair& t = vp[d & 1];
+        if (d + 1 < DB)
+          attn::tr_pair_wait<4>(&t);
+        else
+          attn::tr_pair_wait<0>(&t);
+        o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[0], t.a, o_acc[d], 0, 0, 0);
+        o_acc[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[1], t.b, o_acc[d], 0, 0, 0);
       }
     }
   }

@@ -157,6 +157,41 @@ ZTA_DEV void tr_quad_wait(TrQuad* q) {
                : "i"(N));
 }
 
+// Pipelined pair form (see tr_quad_issue/tr_quad_wait for the contract).
+ZTA_DEV void tr_pair_issue(const uint16_t* lds, int k0, int j0, TrPair* out) {
+  const int l = threadIdx.x & 63;
+  const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
+  const int rb = 8 * (l >> 5) + ((l >> 2) & 3);
+  int a[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int row = k0 + 16 * (i >> 1) + 4 * (i & 1) + rb;
+    a[i] = (int)(size_t)((const char*)lds + row * 256 + (colb ^ ((row & 7) << 4)));
+  }
+  union U {
+    i32x2 d[4];
+    TrPair f;
+  }* u = (union U*)out;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n\t"
+      "ds_read_b64_tr_b16 %1, %5\n\t"
+      "ds_read_b64_tr_b16 %2, %6\n\t"
+      "ds_read_b64_tr_b16 %3, %7"
+      : "=&v"(u->d[0]), "=&v"(u->d[1]), "=&v"(u->d[2]), "=&v"(u->d[3])
+      : "v"(a[0]), "v"(a[1]), "v"(a[2]), "v"(a[3]));
+}
+
+template <int N>
+ZTA_DEV void tr_pair_wait(TrPair* q) {
+  union U {
+    i32x2 d[4];
+    TrPair f;
+  }* u = (union U*)q;
+  asm volatile("s_waitcnt lgkmcnt(%4)"
+               : "+v"(u->d[0]), "+v"(u->d[1]), "+v"(u->d[2]), "+v"(u->d[3])
+               : "i"(N));
+}
+
 ZTA_DEV TrPair tr_frag_pair(const uint16_t* lds, int k0, int j0) {
   const int l = threadIdx.x & 63;
   const int colb = (j0 + (l & 16) + 4 * (l & 3)) * 2;
