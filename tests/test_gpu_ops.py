@@ -340,3 +340,31 @@ def test_inference_kv_cache_gpu(dev):
     assert torch.allclose(step[:, -1], full[:, -1], atol=1e-3, rtol=1e-3)
     out = model.generate(idx, max_new_tokens=8)
     assert out.shape == (2, 24)
+
+
+def test_attention_decode_kernel(dev):
+    """Decode kernel vs SDPA with the explicit ALiBi mask (bf16 and fp16)."""
+    from zero_transformer_amd import ops as O
+    from zero_transformer_amd.ops import reference
+    from zero_transformer_amd.models.inference import _alibi_bias
+
+    torch.manual_seed(13)
+    for dtype in (torch.bfloat16, torch.float16):
+        B, H, S, D = 2, 4, 300, 128
+        q = torch.randn(B, H, 1, D, device=dev).to(dtype)
+        k = torch.randn(B, H, S, D, device=dev).to(dtype)
+        v = torch.randn(B, H, S, D, device=dev).to(dtype)
+        slopes = reference.alibi_slopes(H).to(dev)
+        got = O.attention_decode(q, k, v, slopes)
+        mask = _alibi_bias(slopes, 1, S, dev, torch.float32).unsqueeze(0)
+        want = torch.nn.functional.scaled_dot_product_attention(
+            q.float(), k.float(), v.float(), attn_mask=mask
+        )
+        diff = (got.float() - want).abs().max().item()
+        assert diff < 2e-2, f"{dtype}: decode max diff {diff}"
+        # no-alibi path
+        got0 = O.attention_decode(q, k, v, None)
+        want0 = torch.nn.functional.scaled_dot_product_attention(
+            q.float(), k.float(), v.float()
+        )
+        assert (got0.float() - want0).abs().max().item() < 2e-2

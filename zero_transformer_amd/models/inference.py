@@ -68,7 +68,19 @@ class InferenceAttention(nn.Module):
             v = torch.cat([pv, v], dim=2)
         present = (k, v) if use_cache else None
         Tk = k.shape[2]
-        if self.slopes is not None:
+        if (
+            T == 1
+            and x.is_cuda
+            and q.dtype in (torch.bfloat16, torch.float16)
+            and D <= 256
+        ):
+            # MI355X-native decode kernel (ops/csrc/attn_decode.hip) — the
+            # torch-SDPA path below stays for prefill / CPU
+            from ..ops import attention_decode
+
+            out = attention_decode(q.contiguous(), k.contiguous(), v.contiguous(),
+                                   self.slopes)
+        elif self.slopes is not None:
             mask = _alibi_bias(self.slopes, T, Tk, x.device, q.dtype).unsqueeze(0)
             out = F.scaled_dot_product_attention(q, k, v, attn_mask=mask)
         else:

@@ -259,6 +259,18 @@ def attention(
     return reference.attention(q, k, v, slopes, dropout_p, training)
 
 
+def attention_decode(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+    slopes: Optional[torch.Tensor],
+) -> torch.Tensor:
+    """Single-token KV-cache attention: q (B,H,1,D), k/v (B,H,S,D) ->
+    (B,H,1,D). ALiBi bias slope*(j - (S-1)); every cached key is
+    causal-valid for the one query. Inference only (no autograd)."""
+    if slopes is None:
+        slopes = torch.zeros(q.shape[1], dtype=torch.float32, device=q.device)
+    return hip_ops().attn_decode(q.contiguous(), k.contiguous(), v.contiguous(), slopes)
+
+
 # ---------------------------------------------------------------------------
 # Fused residual-add + dropout:  y = x + dropout(h, p)
 # ---------------------------------------------------------------------------

@@ -28,6 +28,7 @@ SOURCES = [
     "adamw.hip",
     "residual.hip",
     "attention_fwd.hip",
+    "attn_decode.hip",
     "attention_bwd.hip",
 ]
 

@@ -1,0 +1,253 @@
+// Single-token (decode) causal ALiBi attention for gfx950: q (B, H, 1, D)
+// against a KV cache k/v (B, H, S, D) -> out (B, H, 1, D).
+//
+// The training kernels tile MFMA over the T^2 score matrix; decode is a
+// bandwidth-bound row problem (read the cache once), so this is a
+// wave-per-key reduction kernel: one block per (b, h); each wave walks keys
+// j = wave, wave+NW, ... computing score_j = q . k_j (lane-split over D,
+// wave-reduced), keeps wave-local online-softmax state (m, l, fp32 out
+// accumulator in registers), and the four waves merge through LDS at the
+// end. fp32 softmax throughout (the reference's inference mirror used
+// F.scaled_dot_product_attention, torch_compatability/GPT2.py:237 — this is
+// its MI355X-native replacement).
+
+#include "common.h"
+
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+#include <torch/extension.h>
+
+namespace {
+
+constexpr int NW = 4;  // waves per block
+constexpr float NEG_INF = -3.0e38f;
+
+template <typename T16>
+ZTA_DEV float ld(const T16* p);
+template <>
+ZTA_DEV float ld<uint16_t>(const uint16_t* p) { return bf16_to_f32(*p); }
+
+// DPL: D per lane (D/64); supports D up to 256.
+template <int DPL>
+__global__ __launch_bounds__(NW * 64) void attn_decode_kernel(
+    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
+    const uint16_t* __restrict__ v, const float* __restrict__ slopes,
+    uint16_t* __restrict__ out, int H, int S, int D, float scale, bool fp16) {
+  __shared__ float red[NW * (3 + 64 * DPL)];  // per-wave {m, l, acc[D]}
+  const int bh = blockIdx.x;
+  const int h = bh % H;
+  const long base = (long)bh * S * D;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const float slope = slopes[h];
+
+  // q for this (b, h): each lane holds q[lane + e*64], e < DPL (fp32)
+  float qv[DPL];
+#pragma unroll
+  for (int e = 0; e < DPL; ++e) {
+    const int d = lane + e * 64;
+    qv[e] = d < D ? bf16_to_f32(q[(long)bh * D + d]) : 0.f;
+  }
+
+  float m = NEG_INF, l = 0.f;
+  float acc[DPL];
+#pragma unroll
+  for (int e = 0; e < DPL; ++e) acc[e] = 0.f;
+
+  for (int j = wave; j < S; j += NW) {
+    const uint16_t* kr = &k[base + (long)j * D];
+    float s = 0.f;
+#pragma unroll
+    for (int e = 0; e < DPL; ++e) {
+      const int d = lane + e * 64;
+      if (d < D) s += qv[e] * bf16_to_f32(kr[d]);
+    }
+    s = wave_reduce_sum(s);
+    s = s * scale + slope * (float)(j - (S - 1));  // ALiBi; all keys causal-valid
+    float alpha = 1.f;
+    if (s > m) {
+      alpha = m > 0.5f * NEG_INF ? __expf(m - s) : 0.f;
+      m = s;
+    }
+    const float p = __expf(s - m);
+    l = l * alpha + p;
+    const uint16_t* vr = &v[base + (long)j * D];
+#pragma unroll
+    for (int e = 0; e < DPL; ++e) {
+      const int d = lane + e * 64;
+      acc[e] = acc[e] * alpha + (d < D ? p * bf16_to_f32(vr[d]) : 0.f);
+    }
+  }
+
+  // merge the NW waves' online states
+  float* wr = &red[wave * (3 + 64 * DPL)];
+  if (lane == 0) {
+    wr[0] = m;
+    wr[1] = l;
+  }
+#pragma unroll
+  for (int e = 0; e < DPL; ++e) wr[3 + lane + e * 64] = acc[e];
+  __syncthreads();
+  if (wave == 0) {
+    float mg = NEG_INF;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) mg = fmaxf(mg, red[w * (3 + 64 * DPL)]);
+    float lg = 0.f;
+    float og[DPL];
+#pragma unroll
+    for (int e = 0; e < DPL; ++e) og[e] = 0.f;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) {
+      const float mw = red[w * (3 + 64 * DPL)];
+      const float f = mw > 0.5f * NEG_INF ? __expf(mw - mg) : 0.f;
+      lg += red[w * (3 + 64 * DPL) + 1] * f;
+#pragma unroll
+      for (int e = 0; e < DPL; ++e)
+        og[e] += red[w * (3 + 64 * DPL) + 3 + lane + e * 64] * f;
+    }
+    const float inv = lg > 0.f ? 1.f / lg : 0.f;
+#pragma unroll
+    for (int e = 0; e < DPL; ++e) {
+      const int d = lane + e * 64;
+      if (d < D) {
+        const float o = og[e] * inv;
+        if (fp16) {
+          __half hv = __float2half(o);
+          out[(long)bh * D + d] = *reinterpret_cast<uint16_t*>(&hv);
+        } else {
+          out[(long)bh * D + d] = f32_to_bf16(o);
+        }
+      }
+    }
+  }
+}
+
+// fp16 cache variant reads via __half
+template <int DPL>
+__global__ __launch_bounds__(NW * 64) void attn_decode_kernel_f16(
+    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
+    const uint16_t* __restrict__ v, const float* __restrict__ slopes,
+    uint16_t* __restrict__ out, int H, int S, int D, float scale) {
+  __shared__ float red[NW * (3 + 64 * DPL)];
+  const int bh = blockIdx.x;
+  const int h = bh % H;
+  const long base = (long)bh * S * D;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const float slope = slopes[h];
+  auto h2f = [](uint16_t u) {
+    __half hv = *reinterpret_cast<__half*>(&u);
+    return __half2float(hv);
+  };
+  float qv[DPL];
+#pragma unroll
+  for (int e = 0; e < DPL; ++e) {
+    const int d = lane + e * 64;
+    qv[e] = d < D ? h2f(q[(long)bh * D + d]) : 0.f;
+  }
+  float m = NEG_INF, l = 0.f;
+  float acc[DPL];
+#pragma unroll
+  for (int e = 0; e < DPL; ++e) acc[e] = 0.f;
+  for (int j = wave; j < S; j += NW) {
+    const uint16_t* kr = &k[base + (long)j * D];
+    float s = 0.f;
+#pragma unroll
+    for (int e = 0; e < DPL; ++e) {
+      const int d = lane + e * 64;
+      if (d < D) s += qv[e] * h2f(kr[d]);
+    }
+    s = wave_reduce_sum(s);
+    s = s * scale + slope * (float)(j - (S - 1));
+    float alpha = 1.f;
+    if (s > m) {
+      alpha = m > 0.5f * NEG_INF ? __expf(m - s) : 0.f;
+      m = s;
+    }
+    const float p = __expf(s - m);
+    l = l * alpha + p;
+    const uint16_t* vr = &v[base + (long)j * D];
+#pragma unroll
+    for (int e = 0; e < DPL; ++e) {
+      const int d = lane + e * 64;
+      acc[e] = acc[e] * alpha + (d < D ? p * h2f(vr[d]) : 0.f);
+    }
+  }
+  float* wr = &red[wave * (3 + 64 * DPL)];
+  if (lane == 0) {
+    wr[0] = m;
+    wr[1] = l;
+  }
+#pragma unroll
+  for (int e = 0; e < DPL; ++e) wr[3 + lane + e * 64] = acc[e];
+  __syncthreads();
+  if (wave == 0) {
+    float mg = NEG_INF;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) mg = fmaxf(mg, red[w * (3 + 64 * DPL)]);
+    float lg = 0.f;
+    float og[DPL];
+#pragma unroll
+    for (int e = 0; e < DPL; ++e) og[e] = 0.f;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) {
+      const float mw = red[w * (3 + 64 * DPL)];
+      const float f = mw > 0.5f * NEG_INF ? __expf(mw - mg) : 0.f;
+      lg += red[w * (3 + 64 * DPL) + 1] * f;
+#pragma unroll
+      for (int e = 0; e < DPL; ++e)
+        og[e] += red[w * (3 + 64 * DPL) + 3 + lane + e * 64] * f;
+    }
+    const float inv = lg > 0.f ? 1.f / lg : 0.f;
+#pragma unroll
+    for (int e = 0; e < DPL; ++e) {
+      const int d = lane + e * 64;
+      if (d < D) {
+        __half hv = __float2half(og[e] * inv);
+        out[(long)bh * D + d] = *reinterpret_cast<uint16_t*>(&hv);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+at::Tensor attn_decode(at::Tensor q, at::Tensor k, at::Tensor v,
+                       at::Tensor slopes) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.dim() == 4 && q.size(2) == 1,
+              "q must be (B, H, 1, D) contiguous");
+  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  const int B = q.size(0), H = q.size(1), D = q.size(3);
+  const int S = k.size(2);
+  TORCH_CHECK(D <= 256, "attn_decode: head_dim up to 256");
+  auto out = at::empty_like(q);
+  auto sl = slopes.to(at::kFloat).contiguous();
+  const float scale = 1.0f / sqrtf((float)D);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int dpl = (D + 63) / 64;
+  const bool f16 = q.scalar_type() == at::kHalf;
+  TORCH_CHECK(f16 || q.scalar_type() == at::kBFloat16, "bf16 or fp16 only");
+#define LAUNCH(DPL)                                                                \
+  if (f16)                                                                         \
+    hipLaunchKernelGGL(attn_decode_kernel_f16<DPL>, dim3(B * H), dim3(NW * 64), 0, \
+                       stream, (const uint16_t*)q.data_ptr(),                      \
+                       (const uint16_t*)k.data_ptr(), (const uint16_t*)v.data_ptr(),\
+                       sl.data_ptr<float>(), (uint16_t*)out.data_ptr(), H, S, D,   \
+                       scale);                                                     \
+  else                                                                             \
+    hipLaunchKernelGGL(attn_decode_kernel<DPL>, dim3(B * H), dim3(NW * 64), 0,     \
+                       stream, (const uint16_t*)q.data_ptr(),                      \
+                       (const uint16_t*)k.data_ptr(), (const uint16_t*)v.data_ptr(),\
+                       sl.data_ptr<float>(), (uint16_t*)out.data_ptr(), H, S, D,   \
+                       scale, false)
+  switch (dpl) {
+    case 1: LAUNCH(1); break;
+    case 2: LAUNCH(2); break;
+    case 3: LAUNCH(3); break;
+    case 4: LAUNCH(4); break;
+    default: TORCH_CHECK(false, "bad head_dim");
+  }
+#undef LAUNCH
+  return out;
+}

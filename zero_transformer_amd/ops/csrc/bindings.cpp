@@ -18,6 +18,7 @@ void adamw_step(at::Tensor p, at::Tensor p_bf16, at::Tensor g, at::Tensor m,
                 double eps, double wd, double clip, double grad_scale);
 at::Tensor residual_dropout_fwd(at::Tensor x, at::Tensor h, double p, int64_t seed);
 at::Tensor residual_dropout_bwd(at::Tensor dy, double p, int64_t seed);
+at::Tensor attn_decode(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor slopes);
 std::vector<at::Tensor> attn_fwd(at::Tensor qkv, at::Tensor slopes, int64_t H,
                                  double p_drop, int64_t seed);
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor qkv, at::Tensor slopes,
@@ -34,6 +35,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step", &adamw_step, "fused ZeRO-1 AdamW shard step (gfx950)");
   m.def("residual_dropout_fwd", &residual_dropout_fwd, "fused x + dropout(h) fwd (gfx950)");
   m.def("residual_dropout_bwd", &residual_dropout_bwd, "fused x + dropout(h) bwd (gfx950)");
+  m.def("attn_decode", &attn_decode, "single-token KV-cache ALiBi attention (gfx950)");
   m.def("attn_fwd", &attn_fwd, "fused causal ALiBi flash attention fwd (gfx950 MFMA)");
   m.def("attn_bwd", &attn_bwd, "fused causal ALiBi flash attention bwd (gfx950 MFMA)");
 }
