@@ -111,6 +111,31 @@ def main():
             )
             return {"completion": text}
 
+        @app.get("/")
+        def index():
+            # minimal demo page (the reference's Gradio UI role)
+            from fastapi.responses import HTMLResponse
+
+            return HTMLResponse(
+                """<!doctype html><title>zero_transformer_amd</title>
+<h2>zero_transformer_amd inference</h2>
+<textarea id=p rows=6 cols=80>Hello</textarea><br>
+max new tokens <input id=n value=128 size=4>
+temperature <input id=t value=0.8 size=4>
+top-p <input id=tp value=0.95 size=4>
+<button onclick="go()">Generate</button>
+<pre id=out></pre>
+<script>
+async function go(){
+  const r = await fetch('/generate', {method:'POST',
+    headers:{'Content-Type':'application/json'},
+    body: JSON.stringify({prompt: p.value, max_new_tokens: +n.value,
+                          temperature: +t.value, top_p: +tp.value})});
+  out.textContent = (await r.json()).completion;
+}
+</script>"""
+            )
+
         uvicorn.run(app, host="0.0.0.0", port=args.port)
     else:
         prompt = args.prompt or "Hello"

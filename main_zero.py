@@ -129,7 +129,19 @@ def main():
         try:
             import wandb
 
-            wandb_run = wandb.init(project=cfg.data.wandb_project, config=flatten_dict(cfg))
+            # persist the run id next to the checkpoints so --resume
+            # continues the same wandb run (reference resumes its run)
+            os.makedirs(workdir, exist_ok=True)
+            idf = os.path.join(workdir, "wandb_run_id")
+            run_id = None
+            if args.resume and os.path.exists(idf):
+                run_id = open(idf).read().strip() or None
+            wandb_run = wandb.init(
+                project=cfg.data.wandb_project, config=flatten_dict(cfg),
+                id=run_id, resume="allow" if run_id else None,
+            )
+            with open(idf, "w") as f:
+                f.write(wandb_run.id)
         except Exception as e:  # no network on GPU boxes
             log.warning("wandb unavailable: %s", e)
 
