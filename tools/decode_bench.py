@@ -22,6 +22,8 @@ def main():
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--tokens", type=int, default=128)
     p.add_argument("--dtype", default="fp16", choices=["fp16", "bf16"])
+    p.add_argument("--fast", action="store_true",
+                   help="static KV cache + hipGraph-captured decode step")
     args = p.parse_args()
     dev = torch.device("cuda", 0)
     torch.manual_seed(0)
@@ -44,9 +46,21 @@ def main():
         torch.cuda.synchronize()
         return (args.tokens - 8) / (time.perf_counter() - t0)
 
+    if args.fast:
+        from zero_transformer_amd.models.inference import generate_fast
+
+        @torch.no_grad()
+        def run():
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            generate_fast(model, idx, args.tokens)
+            torch.cuda.synchronize()
+            return args.tokens / (time.perf_counter() - t0)
+
     run()  # warmup pass
     tps = run()
-    print(f"{args.model_size} {args.dtype} batch {args.batch}: "
+    mode = "fast(graph)" if args.fast else "dynamic"
+    print(f"{args.model_size} {args.dtype} batch {args.batch} {mode}: "
           f"{tps * args.batch:.1f} tokens/s ({1e3 / tps:.2f} ms/token)")
 
 

@@ -22,6 +22,28 @@ from ..ops.reference import alibi_slopes
 from ..utils.config import load_config
 
 
+class StaticKVCache:
+    """Preallocated per-layer KV buffers with a DEVICE-side length counter.
+
+    Makes the whole decode step hipGraph-replayable: the cache write position
+    and the decode-attention kernel's live length both come from device
+    memory (ops.attention_decode reads `len_t` in-kernel), so one captured
+    graph serves every token as the cache grows.
+    """
+
+    def __init__(self, n_layers: int, batch: int, heads: int, head_dim: int,
+                 max_ctx: int, device, dtype):
+        self.k = [torch.zeros(batch, heads, max_ctx, head_dim, device=device,
+                              dtype=dtype) for _ in range(n_layers)]
+        self.v = [torch.zeros(batch, heads, max_ctx, head_dim, device=device,
+                              dtype=dtype) for _ in range(n_layers)]
+        self.len_t = torch.zeros(1, dtype=torch.int32, device=device)
+        self.max_ctx = max_ctx
+
+    def advance(self, n: int = 1):
+        self.len_t += n  # device add: correct inside a captured graph replay
+
+
 def _alibi_bias(slopes: torch.Tensor, Tq: int, Tk: int, device, dtype) -> torch.Tensor:
     """Additive (H, Tq, Tk) mask: ALiBi bias + causal -inf.
 
@@ -56,12 +78,33 @@ class InferenceAttention(nn.Module):
         x: torch.Tensor,
         layer_past: Optional[Tuple[torch.Tensor, torch.Tensor]] = None,
         use_cache: bool = False,
+        static_cache: Optional["StaticKVCache"] = None,
+        layer_idx: int = 0,
     ):
         B, T, C = x.shape
         H, D = self.num_head, self.head_dim
         q = self.query(x).view(B, T, H, D).transpose(1, 2)
         k = self.key(x).view(B, T, H, D).transpose(1, 2)
         v = self.value(x).view(B, T, H, D).transpose(1, 2)
+        if static_cache is not None:
+            kc, vc = static_cache.k[layer_idx], static_cache.v[layer_idx]
+            if T == 1:
+                # decode: write at device position len-1 (len already advanced
+                # for this token), attend over the live cache via the native
+                # decode kernel — fully graph-replayable
+                pos = (static_cache.len_t.to(torch.long) - 1)
+                kc.index_copy_(2, pos, k.to(kc.dtype))
+                vc.index_copy_(2, pos, v.to(vc.dtype))
+                from ..ops import attention_decode
+
+                out = attention_decode(q.contiguous(), kc, vc, self.slopes,
+                                       s_used=static_cache.len_t)
+                out = out.transpose(1, 2).reshape(B, T, C)
+                return self.fc_resid(out), None
+            # prefill: fill rows [0, T) and fall through to the SDPA path
+            idxs = torch.arange(T, device=x.device)
+            kc.index_copy_(2, idxs, k.to(kc.dtype))
+            vc.index_copy_(2, idxs, v.to(vc.dtype))
         if layer_past is not None:
             pk, pv = layer_past
             k = torch.cat([pk, k], dim=2)  # concat along seq (GPT2.py:177-182)
@@ -115,8 +158,10 @@ class InferenceBlock(nn.Module):
         self.ln2 = nn.LayerNorm(dim, elementwise_affine=True, bias=False, eps=1e-6)
         self.mlp = InferenceMLP(dim)
 
-    def forward(self, x, layer_past=None, use_cache=False):
-        a, present = self.attn(self.ln1(x), layer_past, use_cache)
+    def forward(self, x, layer_past=None, use_cache=False, static_cache=None,
+                layer_idx=0):
+        a, present = self.attn(self.ln1(x), layer_past, use_cache,
+                               static_cache=static_cache, layer_idx=layer_idx)
         x = x + a
         x = x + self.mlp(self.ln2(x))
         return x, present
@@ -145,13 +190,15 @@ class GPT2(nn.Module):
         labels: Optional[torch.Tensor] = None,
         use_cache: bool = False,
         past_states: Optional[List] = None,
+        static_cache: Optional[StaticKVCache] = None,
     ):
         x = self.wte(idx)
         presents = [] if use_cache else None
         if past_states is None:
             past_states = [None] * self.N
-        for block, past in zip(self.blocks, past_states):
-            x, present = block(x, past, use_cache)
+        for i, (block, past) in enumerate(zip(self.blocks, past_states)):
+            x, present = block(x, past, use_cache, static_cache=static_cache,
+                               layer_idx=i)
             if use_cache:
                 presents.append(present)
         x = self.norm(x)
@@ -186,6 +233,63 @@ class GPT2(nn.Module):
             nxt = torch.multinomial(probs, 1) if sample else probs.argmax(-1, keepdim=True)
             idx = torch.cat([idx, nxt], dim=1)
         return idx
+
+
+@torch.no_grad()
+def generate_fast(
+    model: "GPT2",
+    idx: torch.Tensor,
+    max_new_tokens: int,
+    use_graph: bool = True,
+) -> torch.Tensor:
+    """Greedy generation with a static KV cache and (optionally) the whole
+    per-token decode step captured in a hipGraph — removes the per-layer
+    Python/launch overhead that dominates single-token latency.
+    """
+    B, T0 = idx.shape
+    dev = idx.device
+    p = next(model.parameters())
+    H = model.blocks[0].attn.num_head
+    D = model.blocks[0].attn.head_dim
+    max_ctx = min(model.num_ctx, T0 + max_new_tokens)
+    cache = StaticKVCache(model.N, B, H, D, max_ctx, dev, p.dtype)
+
+    # prefill
+    logits = model(idx, static_cache=cache)
+    cache.len_t.fill_(T0)
+    cur = logits[:, -1:].argmax(-1)  # (B, 1) — static input buffer
+    out_tokens = [cur.clone()]
+
+    def step():
+        cache.advance(1)
+        return model(cur, static_cache=cache)
+
+    graph = None
+    if use_graph and dev.type == "cuda":
+        warm = torch.cuda.Stream()
+        warm.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(warm):
+            for _ in range(2):
+                step()
+        torch.cuda.current_stream().wait_stream(warm)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            logits_buf = step()
+        # warmup + capture advanced/wrote 3 fake tokens at rows T0..T0+2;
+        # rewind: real tokens overwrite those rows as len re-advances
+        cache.len_t.fill_(T0)
+
+    for i in range(max_new_tokens - 1):
+        if cache.len_t is not None and T0 + 1 + i >= max_ctx:
+            break
+        if graph is not None:
+            graph.replay()
+            nxt = logits_buf[:, -1:].argmax(-1)
+        else:
+            nxt = step()[:, -1:].argmax(-1)
+        cur.copy_(nxt)
+        out_tokens.append(nxt.clone())
+    return torch.cat([idx] + out_tokens, dim=1)
 
 
 def model_getter(

@@ -262,13 +262,22 @@ def attention(
 def attention_decode(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     slopes: Optional[torch.Tensor],
+    s_used: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Single-token KV-cache attention: q (B,H,1,D), k/v (B,H,S,D) ->
     (B,H,1,D). ALiBi bias slope*(j - (S-1)); every cached key is
-    causal-valid for the one query. Inference only (no autograd)."""
+    causal-valid for the one query. Inference only (no autograd).
+
+    `s_used`: optional 1-element int32 CUDA tensor holding the live cache
+    length when k/v are larger preallocated buffers — read in-kernel, so
+    the launch replays correctly from a hipGraph as the cache grows."""
     if slopes is None:
         slopes = torch.zeros(q.shape[1], dtype=torch.float32, device=q.device)
-    return hip_ops().attn_decode(q.contiguous(), k.contiguous(), v.contiguous(), slopes)
+    if s_used is None:
+        s_used = torch.empty(0, dtype=torch.int32, device=q.device)
+    return hip_ops().attn_decode(
+        q.contiguous(), k.contiguous(), v.contiguous(), slopes, s_used
+    )
 
 
 # ---------------------------------------------------------------------------

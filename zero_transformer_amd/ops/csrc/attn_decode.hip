@@ -29,15 +29,20 @@ template <>
 ZTA_DEV float ld<uint16_t>(const uint16_t* p) { return bf16_to_f32(*p); }
 
 // DPL: D per lane (D/64); supports D up to 256.
+// S_alloc: allocated cache rows per (b,h) (the base stride); s_used: device
+// pointer to the LIVE number of cached keys — read in-kernel so the launch
+// is hipGraph-replayable with a growing cache.
 template <int DPL>
 __global__ __launch_bounds__(NW * 64) void attn_decode_kernel(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const float* __restrict__ slopes,
-    uint16_t* __restrict__ out, int H, int S, int D, float scale, bool fp16) {
+    uint16_t* __restrict__ out, int H, int S_alloc, const int* __restrict__ s_used,
+    int D, float scale, bool fp16) {
+  const int S = s_used ? *s_used : S_alloc;
   __shared__ float red[NW * (3 + 64 * DPL)];  // per-wave {m, l, acc[D]}
   const int bh = blockIdx.x;
   const int h = bh % H;
-  const long base = (long)bh * S * D;
+  const long base = (long)bh * S_alloc * D;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const float slope = slopes[h];
@@ -128,11 +133,13 @@ template <int DPL>
 __global__ __launch_bounds__(NW * 64) void attn_decode_kernel_f16(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const float* __restrict__ slopes,
-    uint16_t* __restrict__ out, int H, int S, int D, float scale) {
+    uint16_t* __restrict__ out, int H, int S_alloc, const int* __restrict__ s_used,
+    int D, float scale) {
+  const int S = s_used ? *s_used : S_alloc;
   __shared__ float red[NW * (3 + 64 * DPL)];
   const int bh = blockIdx.x;
   const int h = bh % H;
-  const long base = (long)bh * S * D;
+  const long base = (long)bh * S_alloc * D;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const float slope = slopes[h];
@@ -213,14 +220,21 @@ __global__ __launch_bounds__(NW * 64) void attn_decode_kernel_f16(
 
 }  // namespace
 
+// `s_used`: optional 1-element int32 CUDA tensor with the live cache length
+// (k/v may be larger preallocated buffers); empty tensor -> use k.size(2).
 at::Tensor attn_decode(at::Tensor q, at::Tensor k, at::Tensor v,
-                       at::Tensor slopes) {
+                       at::Tensor slopes, at::Tensor s_used) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.dim() == 4 && q.size(2) == 1,
               "q must be (B, H, 1, D) contiguous");
   TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
   const int B = q.size(0), H = q.size(1), D = q.size(3);
-  const int S = k.size(2);
+  const int S_alloc = k.size(2);
   TORCH_CHECK(D <= 256, "attn_decode: head_dim up to 256");
+  const int* sp = nullptr;
+  if (s_used.numel() > 0) {
+    TORCH_CHECK(s_used.is_cuda() && s_used.scalar_type() == at::kInt);
+    sp = s_used.data_ptr<int>();
+  }
   auto out = at::empty_like(q);
   auto sl = slopes.to(at::kFloat).contiguous();
   const float scale = 1.0f / sqrtf((float)D);
@@ -233,14 +247,14 @@ at::Tensor attn_decode(at::Tensor q, at::Tensor k, at::Tensor v,
     hipLaunchKernelGGL(attn_decode_kernel_f16<DPL>, dim3(B * H), dim3(NW * 64), 0, \
                        stream, (const uint16_t*)q.data_ptr(),                      \
                        (const uint16_t*)k.data_ptr(), (const uint16_t*)v.data_ptr(),\
-                       sl.data_ptr<float>(), (uint16_t*)out.data_ptr(), H, S, D,   \
-                       scale);                                                     \
+                       sl.data_ptr<float>(), (uint16_t*)out.data_ptr(), H, S_alloc,\
+                       sp, D, scale);                                              \
   else                                                                             \
     hipLaunchKernelGGL(attn_decode_kernel<DPL>, dim3(B * H), dim3(NW * 64), 0,     \
                        stream, (const uint16_t*)q.data_ptr(),                      \
                        (const uint16_t*)k.data_ptr(), (const uint16_t*)v.data_ptr(),\
-                       sl.data_ptr<float>(), (uint16_t*)out.data_ptr(), H, S, D,   \
-                       scale, false)
+                       sl.data_ptr<float>(), (uint16_t*)out.data_ptr(), H, S_alloc,\
+                       sp, D, scale, false)
   switch (dpl) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;

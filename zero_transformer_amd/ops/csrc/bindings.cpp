@@ -18,7 +18,8 @@ void adamw_step(at::Tensor p, at::Tensor p_bf16, at::Tensor g, at::Tensor m,
                 double eps, double wd, double clip, double grad_scale);
 at::Tensor residual_dropout_fwd(at::Tensor x, at::Tensor h, double p, int64_t seed);
 at::Tensor residual_dropout_bwd(at::Tensor dy, double p, int64_t seed);
-at::Tensor attn_decode(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor slopes);
+at::Tensor attn_decode(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor slopes,
+                       at::Tensor s_used);
 std::vector<at::Tensor> attn_fwd(at::Tensor qkv, at::Tensor slopes, int64_t H,
                                  double p_drop, int64_t seed);
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor qkv, at::Tensor slopes,
