@@ -44,6 +44,10 @@ def load_tokenizer():
 
 def model_creator(size: str, checkpoint: str | None, device: torch.device):
     """reference app.py:30-39: build -> device -> half -> eval."""
+    if device.type == "cuda":
+        from zero_transformer_amd.utils import gemm_tune
+
+        gemm_tune.enable()
     model = model_getter(size, model_checkpoint=checkpoint)
     model = model.to(device)
     if device.type == "cuda":

@@ -13,6 +13,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 from zero_transformer_amd.models.inference import model_getter
+from zero_transformer_amd.utils import gemm_tune
 
 
 def main():
@@ -25,6 +26,7 @@ def main():
     p.add_argument("--fast", action="store_true",
                    help="static KV cache + hipGraph-captured decode step")
     args = p.parse_args()
+    gemm_tune.enable()  # committed hipBLASLt tunings (incl. decode shapes)
     dev = torch.device("cuda", 0)
     torch.manual_seed(0)
     model = model_getter(args.model_size).to(dev)
