@@ -24,7 +24,6 @@ import torch
 import torch.distributed as dist
 
 from zero_transformer_amd.models import model_getter
-from zero_transformer_amd.parallel import comm
 from zero_transformer_amd.parallel.zero import ZeRO1Optimizer
 from zero_transformer_amd.training.trainer import TrainEngine
 from zero_transformer_amd.utils import checkpoint as ckpt

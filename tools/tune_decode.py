@@ -1,4 +1,6 @@
-import os, sys, torch
+import sys
+
+import torch
 sys.path.insert(0, "/root/repo")
 from zero_transformer_amd.utils import gemm_tune
 assert gemm_tune.enable(tuning=True)

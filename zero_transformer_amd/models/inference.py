@@ -11,7 +11,6 @@ loss. Loads / saves the .pth state-dict contract
 
 from __future__ import annotations
 
-import math
 from typing import List, Optional, Tuple
 
 import torch

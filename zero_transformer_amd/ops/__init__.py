@@ -8,7 +8,6 @@ use the fp32 reference implementations in ops/reference.py.
 
 from __future__ import annotations
 
-import os
 from typing import Optional
 
 import torch

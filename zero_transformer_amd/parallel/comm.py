@@ -13,8 +13,6 @@ becomes an explicit all-gather of updated params.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 

@@ -27,8 +27,7 @@ p2p links carries a parallel chunk of a ring reduce-scatter.
 
 from __future__ import annotations
 
-import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Callable, Dict, List, Optional, Sequence, Tuple
 
 import torch

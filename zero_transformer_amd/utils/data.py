@@ -21,7 +21,7 @@ from __future__ import annotations
 import io
 import os
 import tarfile
-from typing import Iterator, List, Optional
+from typing import Iterator
 
 import numpy as np
 import torch
